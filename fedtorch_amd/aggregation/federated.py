@@ -1,0 +1,238 @@
+# -*- coding: utf-8 -*-
+"""Federated aggregation on flat arenas over RCCL/xGMI.
+
+Reference semantics (star topology: per-param gather -> sum at rank 0 ->
+broadcast, `comms/algorithms/federated/*.py`) are reproduced with weighted
+world collectives: every rank pre-scales its model diff by its rank weight
+(`fedavg.py:17-27`: 1/K_online for online ranks, 0 for the offline server,
+lambda_i*n/K under DRFA/AFL; `qsparse.py:23`: sample-proportional), offline
+ranks contribute zeros, and ONE all-reduce (or fixed-k all-gather for the
+compressed paths) replaces the per-parameter gather+broadcast pair.  Every
+rank applies the identical aggregate, so every rank's server copy stays in
+sync; a single arena broadcast at round start keeps bitwise consistency.
+
+Divergences from the reference (deliberate, documented):
+* distributed SCAFFOLD uses the *gathered* aggregate for the server control
+  update — the reference overwrites it with the local tensor
+  (`scaffold.py:59-64`, a bug) — and divides by n_nodes, not the dataloader
+  --num_workers flag (`scaffold.py:66`).
+* error-feedback memory is only updated on ranks that actually participated
+  (weight != 0): the reference divides by a zero rank_weight for the offline
+  server (`qsparse.py:57`), poisoning its memory with NaNs.
+* top-k compression selects over the WHOLE arena with the same total budget
+  k = numel*ratio/2 (the reference selects per parameter tensor,
+  `flow_utils.py:218-230`); global selection at equal budget keeps strictly
+  more of the largest entries.
+"""
+import numpy as np
+import torch
+
+from fedtorch_amd import ops
+
+
+def rank_weight(args, online_clients, lambda_weight=None,
+                sample_proportional=False):
+    """(weight, num_online, participating) — reference `fedavg.py:17-27`."""
+    num_online = len(online_clients) if 0 in online_clients \
+        else len(online_clients) + 1
+    rank = args.graph.rank
+    if rank not in online_clients:
+        # offline: the server (rank 0) joins with weight 0 (`fedavg.py:19-20`)
+        # and fully-offline ranks never contribute.
+        return 0.0, num_online, False
+    if lambda_weight is not None:
+        w = lambda_weight * args.graph.n_nodes / num_online
+    elif sample_proportional:
+        w = args.num_samples_per_epoch / args.train_dataset_size
+    else:
+        w = 1.0 / num_online
+    return float(w), num_online, True
+
+
+def _buf(work, name, like):
+    if name not in work:
+        work[name] = torch.zeros_like(like)
+    return work[name]
+
+
+def distribute_model_server(comm, server_flat, src=0):
+    """ONE arena broadcast (reference loops P params,
+    `federated/misc.py:22-27`)."""
+    comm.broadcast(server_flat, src=src)
+    return server_flat
+
+
+def distribute_model_server_control(comm, server_flat, ctrl_flat, work,
+                                    src=0):
+    """[model ‖ control] in one 2N broadcast (reference `scaffold.py:76-91`
+    stacks per-param pairs)."""
+    pair = _buf(work, 'pair2n', torch.cat([server_flat, ctrl_flat]))
+    n = server_flat.numel()
+    pair[:n].copy_(server_flat)
+    pair[n:].copy_(ctrl_flat)
+    comm.broadcast(pair, src=src)
+    server_flat.copy_(pair[:n])
+    ctrl_flat.copy_(pair[n:])
+    return server_flat, ctrl_flat
+
+
+def _communicate_dense(args, comm, agg):
+    """dense or adaptive-quantized aggregate of the pre-scaled diffs."""
+    if args.quantized:
+        q, info = ops.quantize(agg, args.quantized_bits)
+        qs = comm.all_gather_flat(q)
+        infos = comm.all_gather_flat(info)
+        ops.dequant_accumulate(qs, infos, agg)
+        # replicate the reference server's lossy re-quantize->broadcast
+        # (`fedavg.py:54-63`): deterministic, identical on every rank.
+        q2, info2 = ops.quantize(agg, args.quantized_bits)
+        agg.copy_(ops.dequantize(q2, info2))
+    else:
+        comm.all_reduce(agg)
+    return agg
+
+
+def _communicate_compressed(args, comm, g, out):
+    """fixed-k top-k all-gather + local decompress-sum."""
+    k = max(int(g.numel() * args.compressed_ratio / 2), 1)
+    v, i = ops.topk_compress(g, k)
+    vs = comm.all_gather_flat(v)
+    idxs = comm.all_gather_flat(i)
+    ops.scatter_accumulate(out, vs, idxs)
+    return out
+
+
+def fedavg_aggregation(args, comm, arena, server_flat, optimizer,
+                       online_clients, lambda_weight=None, work=None):
+    """FedAvg/FedProx/FedAdam sync (reference `federated/fedavg.py:11-98`)."""
+    work = work if work is not None else {}
+    w, _, _ = rank_weight(args, online_clients, lambda_weight)
+    agg = _buf(work, 'agg', arena.flat)
+    ops.weighted_diff_restore(server_flat, arena.flat, agg, w)
+    _communicate_dense(args, comm, agg)
+    if args.federated_type == 'fedadam':
+        _fedadam_normalize(args, arena, agg)
+    optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
+                   apply_in_momentum=False,
+                   apply_out_momentum=args.out_momentum, grad=agg)
+    server_flat.copy_(arena.flat)
+    return server_flat
+
+
+def _fedadam_normalize(args, arena, agg):
+    """FedAdam server update (reference `fedavg.py:81-85`, after
+    arXiv:2003.00295): per-parameter-tensor v_i = beta*v_i + (1-beta)*||g_i||,
+    g_i /= (sqrt(v_i)+tau).  The reference references `np` without importing
+    it (crash); implemented correctly here."""
+    views = arena.views_of(agg)
+    for i, g in enumerate(views):
+        args.fedadam_v[i] = args.fedadam_beta * args.fedadam_v[i] + \
+            (1 - args.fedadam_beta) * float(torch.norm(g))
+        g /= (np.sqrt(args.fedadam_v[i]) + args.fedadam_tau)
+
+
+def fedgate_aggregation(args, comm, arena, server_flat, delta_flat,
+                        memory_flat, optimizer, online_clients, lr,
+                        local_steps, lambda_weight=None, work=None):
+    """FedGATE / FedCOMGATE sync (reference `federated/fedgate.py:14-118`)."""
+    work = work if work is not None else {}
+    w, _, participating = rank_weight(args, online_clients, lambda_weight)
+    agg = _buf(work, 'agg', arena.flat)
+    ops.scaled_diff(server_flat, arena.flat, agg, w)  # no restore yet
+    if args.quantized:
+        _communicate_dense(args, comm, agg)
+        d = agg
+    elif args.compressed:
+        g = _buf(work, 'g', arena.flat)
+        torch.add(agg, memory_flat, alpha=w, out=g)
+        d = _buf(work, 'd', arena.flat)
+        _communicate_compressed(args, comm, g, d)
+        if participating and w != 0:
+            # mem += agg/w - d (reference `fedgate.py:81`)
+            ops.error_feedback_update(memory_flat, agg, d, 1.0 / w)
+    else:
+        comm.all_reduce(agg)
+        d = agg
+    if participating and w != 0 and local_steps > 0:
+        # delta += (server - d - client)/(lr*tau) (reference `fedgate.py:104`)
+        ops.delta_update(delta_flat, server_flat, d, arena.flat,
+                         1.0 / (lr * local_steps))
+    arena.flat.copy_(server_flat)  # restore
+    optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
+                   apply_in_momentum=False,
+                   apply_out_momentum=args.out_momentum, grad=d)
+    server_flat.copy_(arena.flat)
+    return server_flat
+
+
+def scaffold_aggregation(args, comm, arena, server_flat, server_ctrl,
+                         client_ctrl, optimizer, online_clients, lr,
+                         local_steps, lambda_weight=None, work=None):
+    """SCAFFOLD sync (reference `federated/scaffold.py:10-74`): one 2N
+    all-reduce carries [weighted diff ‖ weighted control delta]."""
+    work = work if work is not None else {}
+    w, _, participating = rank_weight(args, online_clients, lambda_weight)
+    pair = _buf(work, 'pair2n_agg', torch.cat([arena.flat, arena.flat]))
+    n = arena.flat.numel()
+    a, b = pair[:n], pair[n:]
+    ctrl_new = None
+    if participating and w != 0 and local_steps > 0:
+        ctrl_new = _buf(work, 'ctrl_new', arena.flat)
+        # c+ = c - c_server + (server - client)/(tau*lr) (`scaffold.py:26-27`)
+        ops.scaffold_control_update(ctrl_new, client_ctrl, server_ctrl,
+                                    server_flat, arena.flat,
+                                    1.0 / (local_steps * lr))
+        ops.scaled_diff(ctrl_new, client_ctrl, b, w)
+    else:
+        b.zero_()
+    ops.weighted_diff_restore(server_flat, arena.flat, a, w)
+    comm.all_reduce(pair)
+    # server control: -= sum_ctrl_delta * |S|/N (`scaffold.py:66`, fixed: N =
+    # n_nodes, and the GATHERED aggregate is used).
+    server_ctrl.sub_(b, alpha=len(online_clients) / args.graph.n_nodes)
+    if ctrl_new is not None:
+        client_ctrl.copy_(ctrl_new)
+    optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
+                   apply_in_momentum=False,
+                   apply_out_momentum=args.out_momentum, grad=a)
+    server_flat.copy_(arena.flat)
+    return server_flat
+
+
+def qsparse_aggregation(args, comm, arena, server_flat, memory_flat,
+                        optimizer, online_clients, lambda_weight=None,
+                        work=None):
+    """Qsparse local-SGD sync (reference `federated/qsparse.py:11-70`):
+    sample-proportional weights, top-k with error feedback."""
+    work = work if work is not None else {}
+    w, _, participating = rank_weight(args, online_clients, lambda_weight,
+                                      sample_proportional=True)
+    agg = _buf(work, 'agg', arena.flat)
+    ops.weighted_diff_restore(server_flat, arena.flat, agg, w)  # early restore
+    g = _buf(work, 'g', arena.flat)
+    torch.add(agg, memory_flat, alpha=w, out=g)
+    d = _buf(work, 'd', arena.flat)
+    _communicate_compressed(args, comm, g, d)
+    if participating and w != 0:
+        ops.error_feedback_update(memory_flat, agg, d, 1.0 / w)
+    optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
+                   apply_in_momentum=False,
+                   apply_out_momentum=args.out_momentum, grad=d)
+    server_flat.copy_(arena.flat)
+    return server_flat
+
+
+def aggregate_models_virtual(args, comm, flat, online_clients, work=None):
+    """Weighted model average via all-reduce (reference
+    `federated/misc.py:39-52`, used for the DRFA kth model)."""
+    work = work if work is not None else {}
+    out = _buf(work, 'virt', flat)
+    if (0 not in online_clients) and (args.graph.rank == 0):
+        w = 0.0
+    elif args.graph.rank in online_clients:
+        w = 1.0 / len(online_clients)
+    else:
+        w = 0.0
+    out.copy_(flat).mul_(w)
+    comm.all_reduce(out)
+    return out

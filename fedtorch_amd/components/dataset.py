@@ -1,0 +1,283 @@
+# -*- coding: utf-8 -*-
+"""Dataset pipeline (parity with reference `components/dataset.py`).
+
+`define_dataset` -> partition -> DataLoader(s); personalization keeps the
+reference splits (80/20 val, 70/20/10 for perfedavg, `dataset.py:168-206`);
+`load_data_batch` does the H2D move (+ timing); `GrowingMinibatchSampler`
+implements the rho-geometric batch growth (`dataset.py:264-317`).
+"""
+import time
+
+import numpy as np
+import torch
+
+from fedtorch_amd.logs.logging import log
+from fedtorch_amd.components.datasets.partition import (
+    DataPartitioner, GrowingBatchPartitioner, FederatedPartitioner)
+from fedtorch_amd.components.datasets.prepare_data import get_dataset
+
+
+def _load_data_batch(args, _input, _target):
+    if 'least_square' in args.arch:
+        _input = _input.float()
+        _target = _target.unsqueeze(1) if _target.dim() == 1 else _target
+        _target = _target.float()
+    elif args.data in ('epsilon', 'url', 'rcv1', 'higgs'):
+        _input, _target = _input.float(), _target.long()
+    if args.graph.on_cuda and torch.cuda.is_available():
+        _input = _input.cuda(non_blocking=True)
+        _target = _target.cuda(non_blocking=True)
+    return _input, _target
+
+
+def load_data_batch(args, _input, _target, tracker):
+    start_data_time = time.time()
+    _input, _target = _load_data_batch(args, _input, _target)
+    end_data_time = time.time()
+    tracker['data_time'].update(end_data_time - start_data_time)
+    tracker['end_data_time'] = end_data_time
+    return _input, _target
+
+
+def define_dataset(args, shuffle, test=True, Partitioner=None,
+                   return_partitioner=False):
+    log('create {} dataset for rank {}'.format(args.data, args.graph.rank),
+        args.debug)
+    train_loader = partition_dataset(
+        args, shuffle, dataset_type='train', Partitioner=Partitioner,
+        return_partitioner=return_partitioner)
+    if return_partitioner:
+        train_loader, Partitioner = train_loader
+    val_loader = val_loader1 = None
+    if args.fed_personal:
+        if args.federated_type == 'perfedavg':
+            train_loader, val_loader, val_loader1 = train_loader
+        else:
+            train_loader, val_loader = train_loader
+    test_loader = partition_dataset(args, shuffle, dataset_type='test') \
+        if test else None
+    get_data_stat(args, train_loader, test_loader)
+    if args.fed_personal:
+        if args.federated_type == 'perfedavg':
+            out = [train_loader, test_loader, val_loader, val_loader1]
+        else:
+            out = [train_loader, test_loader, val_loader]
+    else:
+        out = [train_loader, test_loader]
+    return (out, Partitioner) if return_partitioner else out
+
+
+def partitioner(args, dataset, shuffle, world_size, partition_type='normal',
+                return_partitioner=False):
+    sizes = [1.0 / world_size] * world_size
+    if partition_type == 'normal':
+        part = DataPartitioner(args, dataset, shuffle, sizes)
+    elif partition_type == 'growing':
+        part = GrowingBatchPartitioner(args, dataset, sizes)
+    elif partition_type == 'noniid':
+        part = FederatedPartitioner(args, dataset, shuffle)
+    else:
+        raise ValueError(partition_type)
+    if return_partitioner:
+        return part.use(args.graph.rank), part
+    return part.use(args.graph.rank)
+
+
+def _make_loader(args, data, batch_size, shuffle, drop_last=False):
+    return torch.utils.data.DataLoader(
+        data, batch_size=batch_size, shuffle=shuffle,
+        num_workers=args.num_workers, pin_memory=args.pin_memory,
+        drop_last=drop_last,
+        persistent_workers=args.num_workers > 0)
+
+
+def partition_dataset(args, shuffle, dataset_type, Partitioner=None,
+                      return_partitioner=False):
+    if Partitioner is None:
+        dataset = get_dataset(args, args.data, args.data_dir,
+                              split=dataset_type)
+    else:
+        dataset = Partitioner.data
+    batch_size = args.batch_size
+    world_size = args.graph.n_nodes
+
+    if args.partition_data and dataset_type == 'train':
+        if args.iid_data:
+            if args.data in ('emnist', 'emnist_full', 'synthetic',
+                             'shakespeare'):
+                raise ValueError('dataset {} has no iid structure'.format(
+                    args.data))
+            pt = 'growing' if args.growing_batch_size else 'normal'
+        else:
+            if args.data not in ('mnist', 'fashion_mnist', 'emnist',
+                                 'emnist_full', 'cifar10', 'cifar100',
+                                 'adult', 'synthetic', 'shakespeare'):
+                raise NotImplementedError(
+                    'non-iid split not implemented for %s' % args.data)
+            if args.growing_batch_size:
+                raise ValueError('Growing minibatch size is not designed for '
+                                 'non-iid data distribution')
+            pt = 'noniid'
+        if Partitioner is None:
+            if return_partitioner:
+                data_to_load, Partitioner = partitioner(
+                    args, dataset, shuffle, world_size, partition_type=pt,
+                    return_partitioner=True)
+            else:
+                data_to_load = partitioner(args, dataset, shuffle, world_size,
+                                           partition_type=pt)
+            log('Make {} data partitions and use the subdata.'.format(pt),
+                args.debug)
+        else:
+            data_to_load = Partitioner.use(args.graph.rank)
+    else:
+        if Partitioner is not None:
+            raise ValueError('Partitioner provided but data partition method '
+                             'is not defined!')
+        data_to_load = dataset
+
+    if dataset_type == 'train':
+        args.train_dataset_size = len(dataset)
+    else:
+        args.val_dataset_size = len(dataset)
+    log('  {} samples for {}; this rank loads {} (rank {}).'.format(
+        len(dataset), dataset_type, len(data_to_load), args.graph.rank),
+        args.debug)
+
+    if args.growing_batch_size and dataset_type == 'train':
+        batch_sampler = GrowingMinibatchSampler(
+            data_source=data_to_load, num_epochs=args.num_epochs,
+            num_iterations=args.num_iterations,
+            base_batch_size=args.base_batch_size,
+            max_batch_size=args.max_batch_size)
+        args.num_epochs = batch_sampler.num_epochs
+        args.num_iterations = batch_sampler.num_iterations
+        args.total_data_size = len(data_to_load)
+        args.num_samples_per_epoch = len(data_to_load) / args.num_epochs
+        data_loader = torch.utils.data.DataLoader(
+            data_to_load, batch_sampler=batch_sampler,
+            num_workers=args.num_workers, pin_memory=args.pin_memory)
+    elif dataset_type == 'train':
+        if args.stop_criteria == 'epoch':
+            args.num_iterations = int(
+                len(data_to_load) * args.num_epochs / batch_size)
+        else:
+            args.num_epochs = int(
+                args.num_iterations * batch_size / len(data_to_load))
+        args.total_data_size = len(data_to_load) * args.num_epochs
+        args.num_samples_per_epoch = len(data_to_load)
+
+        if args.fed_personal:
+            data_to_load_val1 = None
+            if args.federated_type == 'perfedavg':
+                val_size = int(0.1 * len(data_to_load))
+                if args.data in ('emnist', 'emnist_full', 'shakespeare'):
+                    data_to_load_train, data_to_load_val1 = \
+                        torch.utils.data.random_split(
+                            data_to_load,
+                            [len(data_to_load) - val_size, val_size])
+                    data_to_load_val = get_dataset(
+                        args, args.data, args.data_dir, split='val')
+                else:
+                    data_to_load_train, data_to_load_val, data_to_load_val1 =\
+                        torch.utils.data.random_split(
+                            data_to_load,
+                            [len(data_to_load) - 3 * val_size, 2 * val_size,
+                             val_size])
+            else:
+                if args.data in ('emnist', 'emnist_full', 'shakespeare'):
+                    data_to_load_train = data_to_load
+                    data_to_load_val = get_dataset(
+                        args, args.data, args.data_dir, split='val')
+                else:
+                    val_size = int(0.2 * len(data_to_load))
+                    data_to_load_train, data_to_load_val = \
+                        torch.utils.data.random_split(
+                            data_to_load,
+                            [len(data_to_load) - val_size, val_size])
+            data_loader = [
+                _make_loader(args, data_to_load_train, batch_size, True),
+                _make_loader(args, data_to_load_val, batch_size, True)]
+            if args.federated_type == 'perfedavg':
+                data_loader = [
+                    data_loader[0],
+                    _make_loader(args, data_to_load_val1, batch_size, True),
+                    data_loader[1]]
+                # reference order is [train, val, val1]; keep (train, val,
+                # val1) via define_dataset unpacking:
+                data_loader = [data_loader[0], data_loader[2], data_loader[1]]
+        else:
+            data_loader = _make_loader(args, data_to_load, batch_size, True)
+    else:
+        data_loader = _make_loader(args, data_to_load, batch_size, False)
+    return (data_loader, Partitioner) if return_partitioner else data_loader
+
+
+def get_data_stat(args, train_loader, test_loader=None):
+    args.num_batches_train_per_device_per_epoch = len(train_loader)
+    args.num_whole_train_batches_per_worker = \
+        args.num_batches_train_per_device_per_epoch * args.num_epochs
+    args.num_warmup_train_batches_per_worker = \
+        args.num_batches_train_per_device_per_epoch * args.lr_warmup_epochs
+    args.num_iterations_per_worker = args.num_iterations
+    args.num_batches_val_per_device_per_epoch = \
+        len(test_loader) if test_loader is not None else 0
+    log('we have {} epochs, {} train batches/device, {} test batches/device, '
+        'batch size {}.'.format(
+            args.num_epochs, args.num_batches_train_per_device_per_epoch,
+            args.num_batches_val_per_device_per_epoch, args.batch_size),
+        args.debug)
+
+
+class GrowingMinibatchSampler(torch.utils.data.Sampler):
+    """rho-geometric growing batch sizes (reference `dataset.py:264-317`)."""
+
+    def __init__(self, data_source, num_epochs=None, num_iterations=None,
+                 base_batch_size=2, rho=1.01, max_batch_size=0):
+        self.data_source = data_source
+        self.base_batch_size = base_batch_size
+        self.rho = rho
+        self.num_samples_per_epoch = len(data_source)
+        self.idx_pool = []
+        self.max_batch_size = max_batch_size
+        if num_epochs is None:
+            if num_iterations is None:
+                raise ValueError('Need one of num_epochs / num_iterations.')
+            self.num_iterations = num_iterations
+            self.num_epochs = int(
+                base_batch_size * (rho ** num_iterations - 1) /
+                ((rho - 1) * self.num_samples_per_epoch)) + 1
+        else:
+            self.num_epochs = num_epochs
+            self.num_iterations = int(
+                np.log(self.num_samples_per_epoch * num_epochs * (rho - 1) /
+                       base_batch_size + 1) / np.log(rho)) + 1
+        for _ in range(self.num_epochs):
+            self.idx_pool.extend(
+                np.random.permutation(self.num_samples_per_epoch).tolist())
+        self.batch_size = [int(base_batch_size * rho ** i) + 1
+                           for i in range(self.num_iterations)]
+        if max_batch_size:
+            b = np.array(self.batch_size)
+            idx = np.flatnonzero(b > max_batch_size)
+            if len(idx) >= 1:
+                n_full = int(np.sum(b[idx]) // max_batch_size)
+                self.batch_size = self.batch_size[:idx[0]] + \
+                    [max_batch_size] * n_full
+                rem = int(np.sum(b[idx]) % max_batch_size)
+                if n_full and rem:
+                    self.batch_size += [rem]
+                self.num_iterations = len(self.batch_size)
+        self.total_num_data = int(np.sum(self.batch_size))
+
+    def __iter__(self):
+        pool = list(self.idx_pool)
+        for bs in self.batch_size:
+            batch = pool[:bs]
+            pool = pool[bs:]
+            if not batch:
+                return
+            yield batch
+
+    def __len__(self):
+        return self.num_iterations

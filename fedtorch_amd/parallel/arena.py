@@ -1,0 +1,121 @@
+# -*- coding: utf-8 -*-
+"""Flat contiguous parameter arena — the central MI355X-native data structure.
+
+The reference iterates ``model.parameters()`` in Python for every hot path
+(aggregation `comms/algorithms/federated/fedavg.py:30-94`, SGD step
+`components/optimizers/sgd.py:81-128`, distribution `federated/misc.py:26`),
+which costs P ~ 65 tiny kernel launches + P tiny messages per model per sync.
+Here every replica's parameters live in ONE contiguous fp32 buffer:
+
+* every optimizer / aggregation hot path is ONE fused HIP kernel launch;
+* every RCCL collective moves ONE arena-sized message over xGMI;
+* auxiliary state (server model, control variates, FedGATE delta/memory,
+  momentum) are plain flat tensors of the same size — no nn.Module clones
+  (the reference deep-copies whole modules for these, `nodes/nodes.py:87-112`).
+
+Parameter tensors are re-pointed (``p.data = view``) into the arena, and
+``p.grad`` is pointed into a twin gradient arena, so autograd accumulates
+directly into the flat buffer — no flatten/unflatten copies anywhere.
+Offsets are aligned to 64 elements (256 B) so vectorized (float4 / short8)
+kernel loads stay aligned per parameter.
+"""
+import torch
+
+ALIGN = 64  # elements; 256 B for fp32 — one LDS bank row on gfx950
+
+
+def _aligned(n, align=ALIGN):
+    return (n + align - 1) // align * align
+
+
+class Arena(object):
+    """Flat fp32 arena backing the parameters (and grads) of a module."""
+
+    def __init__(self, module, device=None, dtype=torch.float32,
+                 with_grads=True, no_decay_predicate=None):
+        """``no_decay_predicate(name, param) -> bool`` marks parameters that
+        weight decay must skip (the reference zeroes wd for params whose name
+        contains 'bn', `components/optimizer.py:8-16`).  Those are packed at
+        the END of the arena so the fused SGD kernel applies wd to the flat
+        prefix ``[0, wd_numel)`` with zero extra memory traffic."""
+        self.module = module
+        named = [(n, p) for n, p in module.named_parameters()
+                 if p.requires_grad]
+        if len(named) == 0:
+            raise ValueError('module has no trainable parameters')
+        if no_decay_predicate is None:
+            no_decay_predicate = lambda name, p: 'bn' in name  # noqa: E731
+        decay = [(n, p) for n, p in named if not no_decay_predicate(n, p)]
+        nodecay = [(n, p) for n, p in named if no_decay_predicate(n, p)]
+        ordered = decay + nodecay
+        self.names = [n for n, _ in ordered]
+        params = [p for _, p in ordered]
+        self.dtype = dtype
+        self.device = device if device is not None else params[0].device
+        self.offsets, self.numels, self.shapes = [], [], []
+        total = 0
+        for p in params:
+            self.offsets.append(total)
+            self.numels.append(p.numel())
+            self.shapes.append(p.shape)
+            total += _aligned(p.numel())
+        # wd applies to flat[:wd_numel] (includes pad gaps, which stay zero).
+        self.wd_numel = (self.offsets[len(decay) - 1] +
+                         _aligned(self.numels[len(decay) - 1])
+                         if decay else 0)
+        if len(nodecay) == 0:
+            self.wd_numel = total
+        self.numel = total
+        # NOTE: the pad gaps stay zero forever: every fused op is linear in
+        # the buffers, so zero gaps stay zero through steps/collectives.
+        self.flat = torch.zeros(total, dtype=dtype, device=self.device)
+        for p, off, n in zip(params, self.offsets, self.numels):
+            self.flat[off:off + n].copy_(p.data.reshape(-1).to(dtype))
+            p.data = self.flat[off:off + n].view(p.shape)
+        self.params = params
+        self.grad = None
+        if with_grads:
+            self.grad = torch.zeros_like(self.flat)
+            self.attach_grads()
+
+    # ---- gradient plumbing -------------------------------------------------
+    def attach_grads(self):
+        """Point every p.grad at its view of the grad arena."""
+        for p, off, n in zip(self.params, self.offsets, self.numels):
+            p.grad = self.grad[off:off + n].view(p.shape)
+
+    def zero_grad(self):
+        self.grad.zero_()
+
+    # ---- flat state helpers ------------------------------------------------
+    def new_buffer(self, zero=True):
+        """A detached flat tensor of arena shape (server copy, control
+        variate, delta, memory, momentum, ...)."""
+        return torch.zeros_like(self.flat) if zero else torch.empty_like(self.flat)
+
+    def clone_flat(self):
+        return self.flat.detach().clone()
+
+    def load_flat(self, buf):
+        """Set model parameters from a flat buffer (single device copy)."""
+        self.flat.copy_(buf)
+
+    def views_of(self, buf):
+        """Per-parameter shaped views of any arena-shaped flat buffer."""
+        return [buf[off:off + n].view(shape)
+                for off, n, shape in zip(self.offsets, self.numels, self.shapes)]
+
+    def check_views(self):
+        """True if every param still points into the arena (a module-level
+        ``p.data = ...`` assignment elsewhere would silently detach it)."""
+        ptr0 = self.flat.data_ptr()
+        end = ptr0 + self.flat.numel() * self.flat.element_size()
+        return all(ptr0 <= p.data_ptr() < end for p in self.params)
+
+    # ---- interop -----------------------------------------------------------
+    def state_dict_flat(self):
+        return {'flat': self.flat}
+
+    def __repr__(self):
+        return 'Arena(n_params={}, numel={}, device={})'.format(
+            len(self.params), self.numel, self.flat.device)

@@ -1,0 +1,140 @@
+# -*- coding: utf-8 -*-
+"""RCCL/xGMI communication layer.
+
+Design vs the reference (`comms/algorithms/federated/*.py`, inventory in
+SURVEY.md §2.4):
+
+* The reference's star topology — per-parameter ``gather`` to rank 0, sum,
+  ``broadcast`` back — becomes ONE weighted all-reduce over the flat arena:
+  each rank pre-scales its diff by its rank weight (offline ranks contribute
+  zero), which is mathematically identical to gather+sum+broadcast of
+  weighted diffs and removes both the server bottleneck and the per-round
+  ``dist.new_group`` churn (xGMI is point-to-point; RCCL's ring all-reduce
+  uses all 7 links where a gather serializes on rank 0's links).
+* Compressed paths use fixed-k all-gather (k identical on every rank), each
+  rank decompress-sums locally.
+* ``comm_time`` accounting (reference wraps every call in ``time.time()``,
+  e.g. `fedavg.py:45-48`) uses hipEvent pairs on GPU so collectives stay
+  async; ``flush_comm_time`` folds elapsed times into ``args.comm_time[-1]``
+  at round end.
+"""
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def dist_ready():
+    return dist.is_available() and dist.is_initialized()
+
+
+class Comm(object):
+    def __init__(self, args):
+        self.args = args
+        self.on_cuda = bool(args.graph.on_cuda and torch.cuda.is_available())
+        self._events = []  # (start, end) hipEvent pairs of this round
+
+    # ---- timing -----------------------------------------------------------
+    def _tic(self):
+        if self.on_cuda:
+            s = torch.cuda.Event(enable_timing=True)
+            e = torch.cuda.Event(enable_timing=True)
+            s.record()
+            return (s, e)
+        return time.time()
+
+    def _toc(self, tok):
+        if self.on_cuda:
+            tok[1].record()
+            self._events.append(tok)
+        else:
+            if self.args.comm_time:
+                self.args.comm_time[-1] += time.time() - tok
+
+    def flush_comm_time(self):
+        """Fold async comm timings into args.comm_time[-1] (ms -> s)."""
+        if self.on_cuda and self._events:
+            torch.cuda.synchronize()
+            total = sum(s.elapsed_time(e) for s, e in self._events) / 1e3
+            if self.args.comm_time:
+                self.args.comm_time[-1] += total
+            self._events = []
+
+    # ---- collectives ------------------------------------------------------
+    def all_reduce(self, flat, op=dist.ReduceOp.SUM):
+        if not dist_ready():
+            return flat
+        tok = self._tic()
+        dist.all_reduce(flat, op=op)
+        self._toc(tok)
+        return flat
+
+    def broadcast(self, flat, src=0):
+        if not dist_ready():
+            return flat
+        tok = self._tic()
+        dist.broadcast(flat, src=src)
+        self._toc(tok)
+        return flat
+
+    def all_gather_flat(self, flat):
+        """All-gather equal-size flat tensors -> [W, N] stacked tensor."""
+        if not dist_ready():
+            return flat.unsqueeze(0)
+        world = dist.get_world_size()
+        out = flat.new_empty((world,) + flat.shape)
+        tok = self._tic()
+        dist.all_gather_into_tensor(out, flat)
+        self._toc(tok)
+        return out
+
+    def barrier(self):
+        if dist_ready():
+            dist.barrier()
+
+    # ---- federated helpers ------------------------------------------------
+    def set_online_clients(self):
+        """Sample the online set on rank 0 and broadcast it (reference
+        `federated/misc.py:10-19`; one small message, world group, no
+        new_group churn)."""
+        import numpy as np
+        args = self.args
+        n = int(args.online_client_rate * len(args.graph.ranks))
+        onl = np.random.permutation(args.graph.ranks)[:n]
+        t = torch.tensor(list(onl), dtype=torch.int32)
+        if dist_ready():
+            tok = self._tic()
+            dist.broadcast(t, src=0)
+            self._toc(tok)
+        return sorted(t.tolist())
+
+    def set_online_clients_drfa(self, lambda_vector):
+        """lambda-weighted sampling without replacement (reference
+        `federated/misc.py:30-37`)."""
+        import numpy as np
+        args = self.args
+        n = int(args.online_client_rate * args.graph.n_nodes)
+        p = lambda_vector.cpu().numpy().astype(np.float64)
+        p = p / p.sum()
+        onl = np.random.choice(np.arange(args.graph.n_nodes), size=n,
+                               replace=False, p=p)
+        t = torch.tensor(list(onl), dtype=torch.int32)
+        if dist_ready():
+            tok = self._tic()
+            dist.broadcast(t, src=0)
+            self._toc(tok)
+        return sorted(t.tolist())
+
+    def gather_scalar(self, value):
+        """All ranks learn everyone's scalar (reference `loss_gather`
+        `misc.py:54-63` gathers to 0; all-gather keeps every rank able to
+        update its dual state without a later broadcast)."""
+        t = torch.tensor([float(value)], dtype=torch.float32)
+        if not dist_ready():
+            return t
+        world = dist.get_world_size()
+        out = t.new_empty(world)
+        tok = self._tic()
+        dist.all_gather_into_tensor(out, t)
+        self._toc(tok)
+        return out
