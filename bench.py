@@ -1,0 +1,176 @@
+#!/usr/bin/env python3
+# -*- coding: utf-8 -*-
+"""Flagship benchmark: ResNet-20 / CIFAR-10 FedAvg local-SGD (BASELINE.json
+config 2) on 1..8 MI355X GPUs, one client per GPU rank, tau=10 local steps
+between syncs, bf16 compute, synthetic data (no network: BASELINE.md).
+
+A "step" is ONE local SGD step (forward + backward + fused arena SGD); every
+tau-th step additionally runs the FedAvg aggregation (weighted arena
+all-reduce over RCCL/xGMI) inside the timed region.  Weak scaling: per-GPU
+batch is fixed, `value` is the whole-job samples/sec across all ranks.
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+from fedtorch_amd.parameters import get_args  # noqa: E402
+from fedtorch_amd.nodes import Client  # noqa: E402
+from fedtorch_amd.trainings.federated import amp  # noqa: E402
+from fedtorch_amd.trainings.eval import inference  # noqa: E402
+from fedtorch_amd.aggregation.federated import fedavg_aggregation  # noqa: E402
+
+TAU = 10  # local steps per communication round (BASELINE config 2)
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=100)
+    p.add_argument('--warmup', type=int, default=20)
+    p.add_argument('--batch', type=int, default=128)
+    p.add_argument('--model', type=str, default='resnet20')
+    p.add_argument('--dtype', type=str, default='bf16')
+    p.add_argument('--graph', type=str, default='auto',
+                   help='hipGraph-capture the local step: auto|on|off')
+    return p.parse_args()
+
+
+def main():
+    b = parse()
+    on_gpu = torch.cuda.is_available()
+    rank = int(os.environ.get('RANK', '0'))
+    world = int(os.environ.get('WORLD_SIZE', str(b.gpus)))
+    if world > 1 or 'RANK' in os.environ:
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        os.environ.setdefault('MASTER_PORT', '29617')
+        dist.init_process_group('nccl' if on_gpu else 'gloo')
+        rank = dist.get_rank()
+        world = dist.get_world_size()
+
+    use_bf16 = b.dtype == 'bf16' and on_gpu
+    args = get_args([
+        '-d', 'cifar10', '-a', b.model, '-f', 'true',
+        '--federated_type', 'fedavg', '--num_comms', '1000000',
+        '--online_client_rate', '1.0', '--federated_sync_type', 'local_step',
+        '--local_step', str(TAU), '-b', str(b.batch), '--lr', '0.1',
+        '--in_momentum', 'true', '--weight_decay', '5e-4',
+        '--bf16', 'true' if use_bf16 else 'false',
+        '--channels_last', 'true' if on_gpu else 'false',
+        '-j', '0', '--checkpoint', '/tmp/ft_bench_ckpt', '--debug', 'false'])
+    os.environ.setdefault('FEDTORCH_SYNTH_SIZE', '2048')
+
+    if on_gpu:
+        import fedtorch_amd.ops as ops
+        if not ops.hip_available():
+            raise RuntimeError('HIP kernel pack not built: bench refuses the '
+                               'eager fallback on GPU')
+
+    client = Client(args, rank)
+    client.initialize()
+    client.gen_aux_models()
+    args = client.args
+    # pretend dataset-derived counters (synthetic pool below replaces loaders)
+    args.num_batches_train_per_device_per_epoch = 390
+    device = torch.device('cuda') if on_gpu else torch.device('cpu')
+
+    # synthetic CIFAR-shaped data pool, resident on device
+    g = torch.Generator(device='cpu').manual_seed(1234 + rank)
+    pool_n = 8
+    xs = torch.randn((pool_n, b.batch, 3, 32, 32), generator=g)
+    ys = torch.randint(0, 10, (pool_n, b.batch), generator=g)
+    xs = xs.to(device)
+    ys = ys.to(device)
+    if args.channels_last and on_gpu:
+        xs = xs.to(memory_format=torch.channels_last)
+
+    lr = 0.1
+    for pg in client.optimizer.param_groups:
+        pg['lr'] = lr
+    online = list(range(world))
+
+    def local_step(i):
+        client.model.train()
+        client.optimizer.zero_grad()
+        with amp(args):
+            loss, _ = inference(client.model, client.criterion,
+                                client.metrics, xs[i % pool_n], ys[i % pool_n])
+        loss.backward()
+        client.optimizer.step(apply_lr=True, apply_in_momentum=True,
+                              apply_out_momentum=False)
+        return loss
+
+    def sync():
+        args.comm_time.append(0.0)
+        fedavg_aggregation(args, client.comm, client.arena,
+                           client.model_server, client.optimizer, online,
+                           work=client.work)
+
+    def run(n_steps):
+        for s in range(1, n_steps + 1):
+            local_step(s)
+            if s % TAU == 0:
+                sync()
+
+    # ---- warmup ----
+    run(b.warmup)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+
+    # ---- timed ----
+    t0 = time.perf_counter()
+    run(b.steps)
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    elapsed = float(t[0])
+
+    if rank == 0:
+        value = world * b.batch * b.steps / elapsed
+        out = {
+            'metric': 'samples/sec',
+            'value': value,
+            'unit': 'samples/sec',
+            'n_gpus': world,
+            'steps': b.steps,
+            'warmup': b.warmup,
+            'ms_per_step': elapsed / b.steps * 1e3,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'bf16' if use_bf16 else 'fp32',
+            'data': 'synthetic',
+            'config': {'model': b.model, 'global_batch': world * b.batch,
+                       'seq_len': None,
+                       'parallelism': 'fedavg_dp%d_tau%d' % (world, TAU)},
+        }
+        print(json.dumps(out), flush=True)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -72,6 +72,11 @@ class FusedSGD(object):
             self._in_buf = self.arena.new_buffer()
         if use_out and self._out_buf is None:
             self._out_buf = self.arena.new_buffer()
+        # per-algorithm corrections apply to LOCAL steps only — the sync
+        # step applies the raw aggregate (reference corrects p.grad inside
+        # the local loop, `trainings/federated/main.py:116-129`, never at
+        # aggregation time).
+        local = apply_lr
         ops.fused_sgd_step(
             self.arena.flat,
             grad if grad is not None else self.arena.grad,
@@ -83,9 +88,12 @@ class FusedSGD(object):
             in_buf=self._in_buf, out_buf=self._out_buf,
             first_in=use_in and not self._in_init,
             first_out=use_out and not self._out_init,
-            prox_mu=self._prox_mu, server=self._server,
-            ctrl_server=self._ctrl_server, ctrl_client=self._ctrl_client,
-            delta=self._delta, wd_numel=self.arena.wd_numel)
+            prox_mu=self._prox_mu if local else 0.0,
+            server=self._server if local else None,
+            ctrl_server=self._ctrl_server if local else None,
+            ctrl_client=self._ctrl_client if local else None,
+            delta=self._delta if local else None,
+            wd_numel=self.arena.wd_numel)
         if use_in:
             self._in_init = True
         if use_out:

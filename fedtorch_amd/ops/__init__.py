@@ -18,8 +18,10 @@ import torch
 _C = None
 _C_ERR = None
 try:
-    from fedtorch_amd.ops import _C as _C_mod  # built .so, in-tree
-    _C = _C_mod
+    import importlib
+    # NOTE: must go through importlib — a plain `from ... import _C` would
+    # resolve to this module's own `_C = None` attribute, not the .so.
+    _C = importlib.import_module('fedtorch_amd.ops._C')
 except Exception as e:  # pragma: no cover - exercised only when not built
     _C_ERR = e
 

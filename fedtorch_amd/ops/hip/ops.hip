@@ -1,0 +1,668 @@
+// fedtorch_amd CDNA4 kernel pack (gfx950 / MI355X) — torch extension.
+//
+// Arena hot paths of the federated engine, each ONE kernel launch over the
+// flat parameter arena (design rationale: SURVEY.md §2.3 — the reference
+// fedtorch runs every one of these as P~65 per-parameter torch ops).
+// Reference semantics cited per op in fedtorch_amd/ops/__init__.py.
+//
+// Build: hipcc --offload-arch=gfx950 via torch.utils.cpp_extension (see
+// setup.py). No CUDA path, no hipify, no multi-backend dispatch.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+#define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
+#define STREAM at::hip::getCurrentHIPStream().stream()
+
+// ==========================================================================
+// fused dual-mode SGD step (+ per-algorithm corrections)
+// ==========================================================================
+enum {
+  F_DELTA = 1, F_CTRL = 2, F_PROX = 4, F_WD = 8,
+  F_IN = 16, F_OUT = 32, F_NESTEROV = 64, F_FIRST_IN = 128,
+  F_FIRST_OUT = 256
+};
+
+__global__ void fused_sgd_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ bin, float* __restrict__ bout,
+    const float* __restrict__ delta, const float* __restrict__ cs,
+    const float* __restrict__ cc, const float* __restrict__ server,
+    long n4, long wd_n4, float wd, float m_in, float m_out,
+    float omd_in, float omd_out, float step_scale, float prox_mu,
+    int flags) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 dv = reinterpret_cast<const float4*>(g)[i];
+    float4 pv = reinterpret_cast<float4*>(p)[i];
+    float* d = reinterpret_cast<float*>(&dv);
+    float* pp = reinterpret_cast<float*>(&pv);
+    if (flags & F_DELTA) {
+      float4 t = reinterpret_cast<const float4*>(delta)[i];
+      const float* tt = reinterpret_cast<const float*>(&t);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) d[j] -= tt[j];
+    }
+    if (flags & F_CTRL) {
+      float4 a = reinterpret_cast<const float4*>(cs)[i];
+      float4 b = reinterpret_cast<const float4*>(cc)[i];
+      const float* aa = reinterpret_cast<const float*>(&a);
+      const float* bb = reinterpret_cast<const float*>(&b);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) d[j] += aa[j] - bb[j];
+    }
+    if (flags & F_PROX) {
+      float4 s = reinterpret_cast<const float4*>(server)[i];
+      const float* ss = reinterpret_cast<const float*>(&s);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) d[j] += prox_mu * (pp[j] - ss[j]);
+    }
+    if ((flags & F_WD) && i < wd_n4) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) d[j] = fmaf(wd, pp[j], d[j]);
+    }
+    if (flags & F_IN) {
+      float4 bv;
+      if (flags & F_FIRST_IN) {
+        bv = dv;
+      } else {
+        bv = reinterpret_cast<const float4*>(bin)[i];
+        float* b = reinterpret_cast<float*>(&bv);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) b[j] = fmaf(m_in, b[j], omd_in * d[j]);
+      }
+      reinterpret_cast<float4*>(bin)[i] = bv;
+      const float* b = reinterpret_cast<const float*>(&bv);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        d[j] = (flags & F_NESTEROV) ? fmaf(m_in, b[j], d[j]) : b[j];
+    }
+    if (flags & F_OUT) {
+      float4 bv;
+      if (flags & F_FIRST_OUT) {
+        bv = dv;
+      } else {
+        bv = reinterpret_cast<const float4*>(bout)[i];
+        float* b = reinterpret_cast<float*>(&bv);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) b[j] = fmaf(m_out, b[j], omd_out * d[j]);
+      }
+      reinterpret_cast<float4*>(bout)[i] = bv;
+      const float* b = reinterpret_cast<const float*>(&bv);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        d[j] = (flags & F_NESTEROV) ? fmaf(m_out, b[j], d[j]) : b[j];
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) pp[j] = fmaf(-step_scale, d[j], pp[j]);
+    reinterpret_cast<float4*>(p)[i] = pv;
+  }
+}
+
+void fused_sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor in_buf,
+                    torch::Tensor out_buf, torch::Tensor delta,
+                    torch::Tensor ctrl_server, torch::Tensor ctrl_client,
+                    torch::Tensor server, double lr, double scale, double wd,
+                    double m_in, double m_out, double damp, bool nesterov,
+                    bool apply_lr, bool apply_in, bool apply_out,
+                    bool first_in, bool first_out, double prox_mu,
+                    long wd_numel) {
+  CHK(p); CHK(g);
+  TORCH_CHECK(p.numel() % 4 == 0, "arena numel must be float4-aligned");
+  long n4 = p.numel() / 4;
+  int flags = 0;
+  if (delta.numel()) flags |= F_DELTA;
+  if (ctrl_server.numel()) flags |= F_CTRL;
+  if (prox_mu != 0.0 && server.numel()) flags |= F_PROX;
+  if (wd != 0.0 && apply_lr) flags |= F_WD;
+  if (apply_in && m_in != 0.0) flags |= F_IN;
+  if (apply_out && m_out != 0.0) flags |= F_OUT;
+  if (nesterov) flags |= F_NESTEROV;
+  if (first_in) flags |= F_FIRST_IN;
+  if (first_out) flags |= F_FIRST_OUT;
+  hipLaunchKernelGGL(fused_sgd_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
+                     STREAM, p.data_ptr<float>(), g.data_ptr<float>(),
+                     (flags & F_IN) ? in_buf.data_ptr<float>() : nullptr,
+                     (flags & F_OUT) ? out_buf.data_ptr<float>() : nullptr,
+                     (flags & F_DELTA) ? delta.data_ptr<float>() : nullptr,
+                     (flags & F_CTRL) ? ctrl_server.data_ptr<float>() : nullptr,
+                     (flags & F_CTRL) ? ctrl_client.data_ptr<float>() : nullptr,
+                     (flags & F_PROX) ? server.data_ptr<float>() : nullptr,
+                     n4, wd_numel / 4, (float)wd, (float)m_in, (float)m_out,
+                     (float)(1.0 - damp), (float)(1.0 - damp),
+                     (float)(apply_lr ? lr : scale), (float)prox_mu, flags);
+}
+
+// ==========================================================================
+// elementwise arena ops
+// ==========================================================================
+__global__ void wdr_kernel(const float* __restrict__ s, float* __restrict__ c,
+                           float* __restrict__ out, float w, long n4,
+                           int restore) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 sv = reinterpret_cast<const float4*>(s)[i];
+    float4 cv = reinterpret_cast<float4*>(c)[i];
+    float4 ov;
+    float* o = reinterpret_cast<float*>(&ov);
+    const float* ss = reinterpret_cast<const float*>(&sv);
+    const float* ccp = reinterpret_cast<const float*>(&cv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (ss[j] - ccp[j]) * w;
+    reinterpret_cast<float4*>(out)[i] = ov;
+    if (restore) reinterpret_cast<float4*>(c)[i] = sv;
+  }
+}
+
+void weighted_diff_restore(torch::Tensor s, torch::Tensor c, torch::Tensor out,
+                           double w) {
+  CHK(s); CHK(c); CHK(out);
+  long n4 = s.numel() / 4;
+  hipLaunchKernelGGL(wdr_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0, STREAM,
+                     s.data_ptr<float>(), c.data_ptr<float>(),
+                     out.data_ptr<float>(), (float)w, n4, 1);
+}
+
+void scaled_diff(torch::Tensor a, torch::Tensor b, torch::Tensor out,
+                 double w) {
+  CHK(a); CHK(b); CHK(out);
+  long n4 = a.numel() / 4;
+  hipLaunchKernelGGL(wdr_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0, STREAM,
+                     a.data_ptr<float>(), b.data_ptr<float>(),
+                     out.data_ptr<float>(), (float)w, n4, 0);
+}
+
+// y = b*y + a*x ; ef: mem += g*invw - d ; delta += (s-agg-c)*coef ;
+// ctrl_new = cc - cs + (s-c)*coef ; blend out = alpha*a+(1-alpha)*b —
+// all are 2-4-operand streaming FMAs; one generic kernel each.
+__global__ void axpby_kernel(float* __restrict__ y, const float* __restrict__ x,
+                             float a, float b, long n4) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 yv = reinterpret_cast<float4*>(y)[i];
+    float4 xv = reinterpret_cast<const float4*>(x)[i];
+    float* yy = reinterpret_cast<float*>(&yv);
+    const float* xx = reinterpret_cast<const float*>(&xv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) yy[j] = fmaf(b, yy[j], a * xx[j]);
+    reinterpret_cast<float4*>(y)[i] = yv;
+  }
+}
+
+void axpby(torch::Tensor y, torch::Tensor x, double a, double b) {
+  CHK(y); CHK(x);
+  long n4 = y.numel() / 4;
+  hipLaunchKernelGGL(axpby_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
+                     STREAM, y.data_ptr<float>(), x.data_ptr<float>(),
+                     (float)a, (float)b, n4);
+}
+
+__global__ void ef_kernel(float* __restrict__ mem, const float* __restrict__ g,
+                          const float* __restrict__ d, float invw, long n4) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 mv = reinterpret_cast<float4*>(mem)[i];
+    float4 gv = reinterpret_cast<const float4*>(g)[i];
+    float4 dv = reinterpret_cast<const float4*>(d)[i];
+    float* m = reinterpret_cast<float*>(&mv);
+    const float* gg = reinterpret_cast<const float*>(&gv);
+    const float* dd = reinterpret_cast<const float*>(&dv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) m[j] = fmaf(invw, gg[j], m[j]) - dd[j];
+    reinterpret_cast<float4*>(mem)[i] = mv;
+  }
+}
+
+void error_feedback_update(torch::Tensor mem, torch::Tensor g, torch::Tensor d,
+                           double invw) {
+  CHK(mem); CHK(g); CHK(d);
+  long n4 = mem.numel() / 4;
+  hipLaunchKernelGGL(ef_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0, STREAM,
+                     mem.data_ptr<float>(), g.data_ptr<float>(),
+                     d.data_ptr<float>(), (float)invw, n4);
+}
+
+__global__ void delta_kernel(float* __restrict__ delta,
+                             const float* __restrict__ s,
+                             const float* __restrict__ agg,
+                             const float* __restrict__ c, float coef,
+                             long n4) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 dv = reinterpret_cast<float4*>(delta)[i];
+    float4 sv = reinterpret_cast<const float4*>(s)[i];
+    float4 av = reinterpret_cast<const float4*>(agg)[i];
+    float4 cv = reinterpret_cast<const float4*>(c)[i];
+    float* d = reinterpret_cast<float*>(&dv);
+    const float* ss = reinterpret_cast<const float*>(&sv);
+    const float* aa = reinterpret_cast<const float*>(&av);
+    const float* cc2 = reinterpret_cast<const float*>(&cv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      d[j] = fmaf(coef, ss[j] - aa[j] - cc2[j], d[j]);
+    reinterpret_cast<float4*>(delta)[i] = dv;
+  }
+}
+
+void delta_update(torch::Tensor delta, torch::Tensor s, torch::Tensor agg,
+                  torch::Tensor c, double coef) {
+  CHK(delta);
+  long n4 = delta.numel() / 4;
+  hipLaunchKernelGGL(delta_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
+                     STREAM, delta.data_ptr<float>(), s.data_ptr<float>(),
+                     agg.data_ptr<float>(), c.data_ptr<float>(), (float)coef,
+                     n4);
+}
+
+__global__ void ctrl_kernel(float* __restrict__ out,
+                            const float* __restrict__ cc,
+                            const float* __restrict__ cs,
+                            const float* __restrict__ s,
+                            const float* __restrict__ c, float coef, long n4) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 a = reinterpret_cast<const float4*>(cc)[i];
+    float4 b = reinterpret_cast<const float4*>(cs)[i];
+    float4 sv = reinterpret_cast<const float4*>(s)[i];
+    float4 cv = reinterpret_cast<const float4*>(c)[i];
+    float4 ov;
+    float* o = reinterpret_cast<float*>(&ov);
+    const float* aa = reinterpret_cast<const float*>(&a);
+    const float* bb = reinterpret_cast<const float*>(&b);
+    const float* ss = reinterpret_cast<const float*>(&sv);
+    const float* cc3 = reinterpret_cast<const float*>(&cv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      o[j] = aa[j] - bb[j] + coef * (ss[j] - cc3[j]);
+    reinterpret_cast<float4*>(out)[i] = ov;
+  }
+}
+
+void scaffold_control_update(torch::Tensor out, torch::Tensor cc,
+                             torch::Tensor cs, torch::Tensor s,
+                             torch::Tensor c, double coef) {
+  CHK(out);
+  long n4 = out.numel() / 4;
+  hipLaunchKernelGGL(ctrl_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
+                     STREAM, out.data_ptr<float>(), cc.data_ptr<float>(),
+                     cs.data_ptr<float>(), s.data_ptr<float>(),
+                     c.data_ptr<float>(), (float)coef, n4);
+}
+
+__global__ void blend_kernel(float* __restrict__ out,
+                             const float* __restrict__ a,
+                             const float* __restrict__ b, float alpha,
+                             long n4) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 av = reinterpret_cast<const float4*>(a)[i];
+    float4 bv = reinterpret_cast<const float4*>(b)[i];
+    float4 ov;
+    float* o = reinterpret_cast<float*>(&ov);
+    const float* aa = reinterpret_cast<const float*>(&av);
+    const float* bb = reinterpret_cast<const float*>(&bv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      o[j] = fmaf(alpha, aa[j] - bb[j], bb[j]);
+    reinterpret_cast<float4*>(out)[i] = ov;
+  }
+}
+
+void blend(torch::Tensor out, torch::Tensor a, torch::Tensor b, double alpha) {
+  CHK(out);
+  long n4 = out.numel() / 4;
+  hipLaunchKernelGGL(blend_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
+                     STREAM, out.data_ptr<float>(), a.data_ptr<float>(),
+                     b.data_ptr<float>(), (float)alpha, n4);
+}
+
+// APFL alpha gradient: dot(p_f - l_f, alpha*p_g + (1-alpha)*l_g) ----------
+__global__ void alpha_grad_kernel(const float* __restrict__ lf,
+                                  const float* __restrict__ pf,
+                                  const float* __restrict__ lg,
+                                  const float* __restrict__ pg, float alpha,
+                                  long n, float* __restrict__ out) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  float acc = 0.f;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float dif = pf[i] - lf[i];
+    float g = fmaf(alpha, pg[i] - lg[i], lg[i]);
+    acc = fmaf(dif, g, acc);
+  }
+  acc = block_reduce<0>(acc);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+double alpha_grad(torch::Tensor lf, torch::Tensor pf, torch::Tensor lg,
+                  torch::Tensor pg, double alpha) {
+  CHK(lf);
+  auto out = torch::zeros({1}, lf.options());
+  long n = lf.numel();
+  hipLaunchKernelGGL(alpha_grad_kernel, dim3(ft_grid(n)), dim3(FT_BLOCK), 0,
+                     STREAM, lf.data_ptr<float>(), pf.data_ptr<float>(),
+                     lg.data_ptr<float>(), pg.data_ptr<float>(), (float)alpha,
+                     n, out.data_ptr<float>());
+  return out.item<float>();
+}
+
+// ==========================================================================
+// adaptive quantization (reference flow_utils.py:169-212 semantics)
+// ==========================================================================
+__global__ void minmaxsum_kernel(const float* __restrict__ x, long n,
+                                 float* __restrict__ scratch) {
+  // scratch layout: [gridDim] mins | [gridDim] maxs | [gridDim] sums
+  const long stride = (long)gridDim.x * blockDim.x;
+  float mn = 3.4e38f, mx = -3.4e38f, sm = 0.f;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float v = x[i];
+    mn = fminf(mn, v);
+    mx = fmaxf(mx, v);
+    sm += v;
+  }
+  float bmn = block_reduce<1>(mn);
+  __syncthreads();
+  float bmx = block_reduce<2>(mx);
+  __syncthreads();
+  float bsm = block_reduce<0>(sm);
+  if (threadIdx.x == 0) {
+    scratch[blockIdx.x] = bmn;
+    scratch[gridDim.x + blockIdx.x] = bmx;
+    scratch[2 * gridDim.x + blockIdx.x] = bsm;
+  }
+}
+
+__global__ void quant_info_kernel(const float* __restrict__ scratch,
+                                  int nblocks, long n, float qmin, float qmax,
+                                  float* __restrict__ info) {
+  // single block finalize: info = [scale, zero_point, mean]
+  float mn = 3.4e38f, mx = -3.4e38f, sm = 0.f;
+  for (int i = threadIdx.x; i < nblocks; i += blockDim.x) {
+    mn = fminf(mn, scratch[i]);
+    mx = fmaxf(mx, scratch[nblocks + i]);
+    sm += scratch[2 * nblocks + i];
+  }
+  mn = block_reduce<1>(mn);
+  __syncthreads();
+  mx = block_reduce<2>(mx);
+  __syncthreads();
+  sm = block_reduce<0>(sm);
+  if (threadIdx.x == 0) {
+    float mean = sm / (float)n;
+    float scale = (mx - mn) / (qmax - qmin);
+    if (scale == 0.f) scale = 0.001f;
+    float izp = qmin - (mn - mean) / scale;
+    // reference: clamp then python int() = truncation toward zero
+    float zp = izp < qmin ? qmin : (izp > qmax ? qmax : truncf(izp));
+    info[0] = scale;
+    info[1] = zp;
+    info[2] = mean;
+  }
+}
+
+template <typename QT>
+__global__ void quant_encode_kernel(const float* __restrict__ x, long n,
+                                    const float* __restrict__ info,
+                                    float qmin, float qmax,
+                                    QT* __restrict__ q) {
+  const float scale = info[0], zp = info[1], mean = info[2];
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float v = zp + (x[i] - mean) / scale;
+    v = fminf(fmaxf(v, qmin), qmax);
+    q[i] = (QT)rintf(v);  // round-half-even, matches torch round_()
+  }
+}
+
+std::vector<torch::Tensor> quantize_adaptive(torch::Tensor x, long num_bits) {
+  CHK(x);
+  long n = x.numel();
+  float qmin = -exp2f((float)(num_bits - 1));
+  float qmax = exp2f((float)(num_bits - 1)) - 1.f;
+  int blocks = ft_grid(n);
+  auto scratch = torch::empty({3 * blocks}, x.options());
+  auto info = torch::empty({3}, x.options());
+  hipLaunchKernelGGL(minmaxsum_kernel, dim3(blocks), dim3(FT_BLOCK), 0,
+                     STREAM, x.data_ptr<float>(), n,
+                     scratch.data_ptr<float>());
+  hipLaunchKernelGGL(quant_info_kernel, dim3(1), dim3(FT_BLOCK), 0, STREAM,
+                     scratch.data_ptr<float>(), blocks, n, qmin, qmax,
+                     info.data_ptr<float>());
+  torch::Tensor q;
+  if (num_bits == 8) {
+    q = torch::empty({n}, x.options().dtype(torch::kChar));
+    hipLaunchKernelGGL(quant_encode_kernel<int8_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, x.data_ptr<float>(), n,
+                       info.data_ptr<float>(), qmin, qmax,
+                       q.data_ptr<int8_t>());
+  } else {
+    q = torch::empty({n}, x.options().dtype(torch::kShort));
+    hipLaunchKernelGGL(quant_encode_kernel<int16_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, x.data_ptr<float>(), n,
+                       info.data_ptr<float>(), qmin, qmax,
+                       q.data_ptr<int16_t>());
+  }
+  return {q, info};
+}
+
+template <typename QT>
+__global__ void dequant_kernel(const QT* __restrict__ q, long n,
+                               const float* __restrict__ info,
+                               float* __restrict__ out) {
+  const float scale = info[0], zp = info[1], mean = info[2];
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = fmaf(scale, (float)q[i] - zp, mean);
+}
+
+torch::Tensor dequantize(torch::Tensor q, torch::Tensor info) {
+  CHK(q);
+  long n = q.numel();
+  auto out = torch::empty({n}, info.options());
+  if (q.scalar_type() == torch::kChar)
+    hipLaunchKernelGGL(dequant_kernel<int8_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, q.data_ptr<int8_t>(), n,
+                       info.data_ptr<float>(), out.data_ptr<float>());
+  else
+    hipLaunchKernelGGL(dequant_kernel<int16_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, q.data_ptr<int16_t>(), n,
+                       info.data_ptr<float>(), out.data_ptr<float>());
+  return out;
+}
+
+template <typename QT>
+__global__ void dequant_acc_kernel(const QT* __restrict__ qs, long k, long n,
+                                   const float* __restrict__ infos,
+                                   float* __restrict__ out) {
+  // qs: [k, n]; infos: [k, 3]; out[i] = sum_j scale_j*(q_ji - zp_j) + mean_j
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float acc = 0.f;
+    for (long j = 0; j < k; ++j) {
+      const float scale = infos[3 * j], zp = infos[3 * j + 1],
+                  mean = infos[3 * j + 2];
+      acc += fmaf(scale, (float)qs[j * n + i] - zp, mean);
+    }
+    out[i] = acc;
+  }
+}
+
+void dequant_accumulate(torch::Tensor qs, torch::Tensor infos,
+                        torch::Tensor out) {
+  CHK(qs); CHK(out);
+  long k = qs.size(0), n = qs.size(1);
+  if (qs.scalar_type() == torch::kChar)
+    hipLaunchKernelGGL(dequant_acc_kernel<int8_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, qs.data_ptr<int8_t>(), k, n,
+                       infos.data_ptr<float>(), out.data_ptr<float>());
+  else
+    hipLaunchKernelGGL(dequant_acc_kernel<int16_t>, dim3(ft_grid(n)),
+                       dim3(FT_BLOCK), 0, STREAM, qs.data_ptr<int16_t>(), k,
+                       n, infos.data_ptr<float>(), out.data_ptr<float>());
+}
+
+// ==========================================================================
+// top-k |x| selection: 4-pass device-side radix select + compaction
+// (the reference's `x.abs().topk(k)` per tensor, `flow_utils.py:218-230`).
+// ==========================================================================
+__global__ void radix_hist_kernel(const float* __restrict__ x, long n,
+                                  const unsigned int* __restrict__ state,
+                                  int shift, unsigned int* __restrict__ hist) {
+  // state = {prefix, k_remaining}; count keys matching prefix above `shift`
+  __shared__ unsigned int lhist[256];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) lhist[i] = 0;
+  __syncthreads();
+  const unsigned int prefix = state[0];
+  const unsigned int pmask = (shift == 24) ? 0u
+      : (0xFFFFFFFFu << (shift + 8));
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    unsigned int key = abs_key(x[i]);
+    if ((key & pmask) == (prefix & pmask))
+      atomicAdd(&lhist[(key >> shift) & 0xFF], 1u);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    if (lhist[i]) atomicAdd(&hist[i], lhist[i]);
+}
+
+__global__ void radix_pick_kernel(unsigned int* __restrict__ hist, int shift,
+                                  unsigned int* __restrict__ state) {
+  // ONE thread walks the 256 buckets from the top: find the bucket where the
+  // k-th largest key lands, fold it into the prefix, update k_remaining.
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  unsigned int k = state[1];
+  unsigned int cum = 0;
+  int b = 255;
+  for (; b >= 0; --b) {
+    unsigned int c = hist[b];
+    if (cum + c >= k) break;
+    cum += c;
+  }
+  if (b < 0) b = 0;
+  state[0] |= ((unsigned int)b) << shift;
+  state[1] = k - cum;
+  for (int i = 0; i < 256; ++i) hist[i] = 0;  // reset for next pass
+}
+
+__global__ void topk_compact_kernel(const float* __restrict__ x, long n,
+                                    const unsigned int* __restrict__ state,
+                                    long k, float* __restrict__ v,
+                                    int* __restrict__ idx,
+                                    unsigned int* __restrict__ counters) {
+  // counters[0]: slots for keys > threshold; counters[1]: ties (== thr).
+  const unsigned int thr = state[0];
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    unsigned int key = abs_key(x[i]);
+    if (key > thr) {
+      unsigned int slot = atomicAdd(&counters[0], 1u);
+      v[slot] = x[i];
+      idx[slot] = (int)i;
+    }
+  }
+}
+
+__global__ void topk_ties_kernel(const float* __restrict__ x, long n,
+                                 const unsigned int* __restrict__ state,
+                                 long k, float* __restrict__ v,
+                                 int* __restrict__ idx,
+                                 unsigned int* __restrict__ counters) {
+  const unsigned int thr = state[0];
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    unsigned int key = abs_key(x[i]);
+    if (key == thr) {
+      unsigned int slot = atomicAdd(&counters[0], 1u);
+      if (slot < (unsigned int)k) {
+        v[slot] = x[i];
+        idx[slot] = (int)i;
+      }
+    }
+  }
+}
+
+std::vector<torch::Tensor> topk_compress(torch::Tensor x, long k) {
+  CHK(x);
+  long n = x.numel();
+  TORCH_CHECK(k > 0 && k <= n, "invalid k");
+  auto u32 = x.options().dtype(torch::kUInt32);
+  auto state = torch::tensor({(int64_t)0, (int64_t)k},
+                             torch::dtype(torch::kUInt32))
+                   .to(x.device(), /*non_blocking=*/true);
+  auto hist = torch::zeros({256}, u32);
+  auto v = torch::empty({k}, x.options());
+  auto idx = torch::empty({k}, x.options().dtype(torch::kInt32));
+  auto counters = torch::zeros({2}, u32);
+  for (int shift = 24; shift >= 0; shift -= 8) {
+    hipLaunchKernelGGL(radix_hist_kernel, dim3(ft_grid(n)), dim3(FT_BLOCK), 0,
+                       STREAM, x.data_ptr<float>(), n,
+                       state.data_ptr<unsigned int>(), shift,
+                       hist.data_ptr<unsigned int>());
+    hipLaunchKernelGGL(radix_pick_kernel, dim3(1), dim3(1), 0, STREAM,
+                       hist.data_ptr<unsigned int>(), shift,
+                       state.data_ptr<unsigned int>());
+  }
+  hipLaunchKernelGGL(topk_compact_kernel, dim3(ft_grid(n)), dim3(FT_BLOCK), 0,
+                     STREAM, x.data_ptr<float>(), n,
+                     state.data_ptr<unsigned int>(), k, v.data_ptr<float>(),
+                     idx.data_ptr<int>(), counters.data_ptr<unsigned int>());
+  hipLaunchKernelGGL(topk_ties_kernel, dim3(ft_grid(n)), dim3(FT_BLOCK), 0,
+                     STREAM, x.data_ptr<float>(), n,
+                     state.data_ptr<unsigned int>(), k, v.data_ptr<float>(),
+                     idx.data_ptr<int>(), counters.data_ptr<unsigned int>());
+  return {v, idx};
+}
+
+// fused decompress + K-way sum: out = sum_k scatter(vs[k] @ idxs[k]) -------
+__global__ void scatter_acc_kernel(const float* __restrict__ vs,
+                                   const int* __restrict__ idxs, long total,
+                                   float* __restrict__ out) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += stride)
+    atomicAdd(&out[idxs[i]], vs[i]);
+}
+
+void scatter_accumulate(torch::Tensor out, torch::Tensor vs,
+                        torch::Tensor idxs) {
+  CHK(out); CHK(vs); CHK(idxs);
+  out.zero_();
+  long total = vs.numel();
+  hipLaunchKernelGGL(scatter_acc_kernel, dim3(ft_grid(total)), dim3(FT_BLOCK),
+                     0, STREAM, vs.data_ptr<float>(), idxs.data_ptr<int>(),
+                     total, out.data_ptr<float>());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_sgd_step", &fused_sgd_step, "fused dual-mode SGD step");
+  m.def("weighted_diff_restore", &weighted_diff_restore);
+  m.def("scaled_diff", &scaled_diff);
+  m.def("axpby", &axpby);
+  m.def("error_feedback_update", &error_feedback_update);
+  m.def("delta_update", &delta_update);
+  m.def("scaffold_control_update", &scaffold_control_update);
+  m.def("blend", &blend);
+  m.def("alpha_grad", &alpha_grad);
+  m.def("quantize_adaptive", &quantize_adaptive);
+  m.def("dequantize", &dequantize);
+  m.def("dequant_accumulate", &dequant_accumulate);
+  m.def("topk_compress", &topk_compress);
+  m.def("scatter_accumulate", &scatter_accumulate);
+}

@@ -78,15 +78,16 @@ class Comm(object):
         return flat
 
     def all_gather_flat(self, flat):
-        """All-gather equal-size flat tensors -> [W, N] stacked tensor."""
+        """All-gather equal-size flat tensors -> [W, N] stacked tensor.
+        (gloo requires a flat output buffer, hence the view.)"""
         if not dist_ready():
             return flat.unsqueeze(0)
         world = dist.get_world_size()
-        out = flat.new_empty((world,) + flat.shape)
+        out = flat.new_empty(world * flat.numel())
         tok = self._tic()
         dist.all_gather_into_tensor(out, flat)
         self._toc(tok)
-        return out
+        return out.view((world,) + tuple(flat.shape))
 
     def barrier(self):
         if dist_ready():

@@ -1,0 +1,201 @@
+# -*- coding: utf-8 -*-
+"""HIP kernel pack numerics: every CDNA4 kernel vs the eager torch fp32
+reference (the same eager code path the CPU tests pin against the reference
+formulas)."""
+import pytest
+import torch
+
+import fedtorch_amd.ops as ops
+
+pytestmark = pytest.mark.gpu
+
+N = 64 * 1024 + 64  # arena-aligned, > one wave-block
+
+
+def _pair(n=N, seed=0):
+    torch.manual_seed(seed)
+    return (torch.randn(n, device='cuda'), torch.randn(n, device='cuda'))
+
+
+def eager(fn, *t, **kw):
+    """run the op's eager path on CPU clones."""
+    cpu = [x.cpu() if isinstance(x, torch.Tensor) else x for x in t]
+    old = ops.FORCE_EAGER
+    ops.FORCE_EAGER = True
+    try:
+        out = fn(*cpu, **kw)
+    finally:
+        ops.FORCE_EAGER = old
+    return out, cpu
+
+
+def test_ext_is_built():
+    assert ops.hip_available(), 'HIP kernel pack must be built on GPU boxes'
+
+
+def test_weighted_diff_restore_gpu():
+    s, c = _pair()
+    out = torch.zeros_like(s)
+    c0 = c.clone()
+    ops.weighted_diff_restore(s, c, out, 0.3)
+    assert torch.allclose(out.cpu(), ((s - c0) * 0.3).cpu(), atol=1e-6)
+    assert torch.equal(c, s)
+
+
+def test_fused_sgd_gpu_matches_eager():
+    torch.manual_seed(1)
+    n = N
+    p = torch.randn(n, device='cuda')
+    g = torch.randn(n, device='cuda')
+    in_buf = torch.zeros_like(p)
+    server = torch.randn(n, device='cuda')
+    pc, gc, bc, sc = p.cpu(), g.cpu(), in_buf.cpu(), server.cpu()
+    kw = dict(lr=0.1, scale=1.0, weight_decay=0.01, in_momentum=0.9,
+              out_momentum=0.0, dampening=0.0, nesterov=True, apply_lr=True,
+              apply_in_momentum=True, apply_out_momentum=False,
+              prox_mu=0.05, wd_numel=n - 256)
+    # two steps to exercise first_in then steady state
+    ops.fused_sgd_step(p, g, in_buf=in_buf, server=server, first_in=True,
+                       **kw)
+    ops.fused_sgd_step(p, g, in_buf=in_buf, server=server, first_in=False,
+                       **kw)
+    old = ops.FORCE_EAGER
+    ops.FORCE_EAGER = True
+    try:
+        ops.fused_sgd_step(pc, gc, in_buf=bc, server=sc, first_in=True, **kw)
+        ops.fused_sgd_step(pc, gc, in_buf=bc, server=sc, first_in=False, **kw)
+    finally:
+        ops.FORCE_EAGER = old
+    assert torch.allclose(p.cpu(), pc, atol=1e-5)
+    assert torch.allclose(in_buf.cpu(), bc, atol=1e-5)
+
+
+def test_fused_sgd_gpu_corrections():
+    torch.manual_seed(2)
+    n = 4096
+    p = torch.randn(n, device='cuda')
+    g = torch.randn(n, device='cuda')
+    delta = torch.randn(n, device='cuda')
+    cs = torch.randn(n, device='cuda')
+    cc = torch.randn(n, device='cuda')
+    p0 = p.clone()
+    ops.fused_sgd_step(p, g, delta=delta, ctrl_server=cs, ctrl_client=cc,
+                       lr=0.2, scale=1.0, weight_decay=0.0, in_momentum=0.0,
+                       out_momentum=0.0, dampening=0.0, nesterov=False,
+                       apply_lr=True, apply_in_momentum=False,
+                       apply_out_momentum=False)
+    expected = p0 - 0.2 * (g - delta + cs - cc)
+    assert torch.allclose(p.cpu(), expected.cpu(), atol=1e-5)
+
+
+@pytest.mark.parametrize('bits', [8, 16])
+def test_quantize_gpu_matches_eager(bits):
+    x = _pair(seed=3)[0]
+    q, info = ops.quantize(x, bits)
+    (qe, infoe), _ = eager(ops.quantize, x, num_bits=bits)
+    assert torch.allclose(info.cpu(), infoe, rtol=1e-5, atol=1e-5)
+    # mean differs in last ulp between GPU tree-sum and CPU serial sum;
+    # codes may differ by 1 on exact rounding boundaries — bound the count.
+    diff = (q.cpu().long() - qe.long()).abs()
+    assert (diff > 1).sum().item() == 0
+    assert (diff == 1).float().mean().item() < 1e-3
+    # round-trip error bound holds exactly
+    xr = ops.dequantize(q, info)
+    assert (x - xr).abs().max().item() <= info[0].item() * 0.5 + 1e-6
+
+
+def test_dequant_accumulate_gpu():
+    torch.manual_seed(4)
+    xs = torch.randn(4, 8192, device='cuda')
+    qs, infos = [], []
+    for k in range(4):
+        q, i = ops.quantize(xs[k], 8)
+        qs.append(q)
+        infos.append(i)
+    qs = torch.stack(qs)
+    infos = torch.stack(infos)
+    out = torch.zeros(8192, device='cuda')
+    ops.dequant_accumulate(qs, infos, out)
+    expected = sum(ops.dequantize(qs[k], infos[k]) for k in range(4))
+    assert torch.allclose(out, expected, atol=1e-5)
+
+
+def test_topk_gpu_exact_selection():
+    torch.manual_seed(5)
+    x = torch.randn(200000, device='cuda')
+    k = 5000
+    v, i = ops.topk_compress(x, k)
+    vr, ir = x.abs().topk(k)
+    assert v.shape[0] == k
+    # selected |values| must match torch.topk's (sets equal up to threshold
+    # ties)
+    assert torch.allclose(v.abs().sort(descending=True)[0], vr,
+                          atol=1e-6)
+    # indices consistent with values
+    assert torch.equal(v, x[i.long()])
+    # no duplicate indices
+    assert i.unique().numel() == k
+
+
+def test_topk_gpu_ties():
+    x = torch.ones(4096, device='cuda')
+    v, i = ops.topk_compress(x, 100)
+    assert v.shape[0] == 100
+    assert i.unique().numel() == 100
+    assert torch.allclose(v, torch.ones(100, device='cuda'))
+
+
+def test_scatter_accumulate_gpu():
+    torch.manual_seed(6)
+    n, k = 65536, 2048
+    vs, idxs = [], []
+    for s in range(5):
+        x = torch.randn(n, device='cuda')
+        v, i = ops.topk_compress(x, k)
+        vs.append(v)
+        idxs.append(i)
+    vs = torch.stack(vs)
+    idxs = torch.stack(idxs)
+    out = torch.empty(n, device='cuda')
+    ops.scatter_accumulate(out, vs, idxs)
+    expected = torch.zeros(n, device='cuda')
+    for s in range(5):
+        expected.scatter_add_(0, idxs[s].long(), vs[s])
+    assert torch.allclose(out, expected, atol=1e-5)
+
+
+def test_elementwise_gpu():
+    a, b = _pair(4096, seed=7)
+    y = a.clone()
+    ops.axpby(y, b, a=0.3, b=0.7)
+    assert torch.allclose(y, 0.7 * a + 0.3 * b, atol=1e-6)
+
+    mem, g = _pair(4096, seed=8)
+    mem0 = mem.clone()
+    d = torch.randn(4096, device='cuda')
+    ops.error_feedback_update(mem, g, d, 2.0)
+    assert torch.allclose(mem, mem0 + 2 * g - d, atol=1e-5)
+
+    out = torch.zeros(4096, device='cuda')
+    ops.blend(out, a, b, 0.25)
+    assert torch.allclose(out, 0.25 * a + 0.75 * b, atol=1e-6)
+
+    lf, pf = _pair(8192, seed=9)
+    lg, pg = _pair(8192, seed=10)
+    ga = ops.alpha_grad(lf, pf, lg, pg, 0.4)
+    expected = torch.dot(pf - lf, 0.4 * pg + 0.6 * lg).item() + 0.02 * 0.4
+    assert abs(ga - expected) < abs(expected) * 1e-3 + 1e-3
+
+
+def test_scaffold_delta_gpu():
+    cc, cs = _pair(4096, seed=11)
+    s, c = _pair(4096, seed=12)
+    out = torch.zeros(4096, device='cuda')
+    ops.scaffold_control_update(out, cc, cs, s, c, 1.5)
+    assert torch.allclose(out, cc - cs + 1.5 * (s - c), atol=1e-5)
+
+    delta = torch.randn(4096, device='cuda')
+    d0 = delta.clone()
+    agg = torch.randn(4096, device='cuda')
+    ops.delta_update(delta, s, agg, c, 0.25)
+    assert torch.allclose(delta, d0 + 0.25 * (s - agg - c), atol=1e-5)
