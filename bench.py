@@ -44,6 +44,8 @@ def parse():
     p.add_argument('--dtype', type=str, default='bf16')
     p.add_argument('--graph', type=str, default='auto',
                    help='hipGraph-capture the local step: auto|on|off')
+    p.add_argument('--layout', type=str, default='nhwc',
+                   choices=['nhwc', 'nchw'])
     return p.parse_args()
 
 
@@ -55,7 +57,7 @@ def main():
     if world > 1 or 'RANK' in os.environ:
         os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
         os.environ.setdefault('MASTER_PORT', '29617')
-        dist.init_process_group('nccl' if on_gpu else 'gloo')
+        dist.init_process_group('cpu:gloo,cuda:nccl' if on_gpu else 'gloo')
         rank = dist.get_rank()
         world = dist.get_world_size()
 
@@ -67,7 +69,8 @@ def main():
         '--local_step', str(TAU), '-b', str(b.batch), '--lr', '0.1',
         '--in_momentum', 'true', '--weight_decay', '5e-4',
         '--bf16', 'true' if use_bf16 else 'false',
-        '--channels_last', 'true' if on_gpu else 'false',
+        '--channels_last',
+        'true' if (on_gpu and b.layout == 'nhwc') else 'false',
         '-j', '0', '--checkpoint', '/tmp/ft_bench_ckpt', '--debug', 'false'])
     os.environ.setdefault('FEDTORCH_SYNTH_SIZE', '2048')
 
@@ -90,26 +93,59 @@ def main():
     pool_n = 8
     xs = torch.randn((pool_n, b.batch, 3, 32, 32), generator=g)
     ys = torch.randint(0, 10, (pool_n, b.batch), generator=g)
-    xs = xs.to(device)
     ys = ys.to(device)
     if args.channels_last and on_gpu:
-        xs = xs.to(memory_format=torch.channels_last)
+        xs = [xs[i].to(device).to(memory_format=torch.channels_last)
+              for i in range(pool_n)]
+    else:
+        xs = xs.to(device)
 
     lr = 0.1
     for pg in client.optimizer.param_groups:
         pg['lr'] = lr
     online = list(range(world))
+    client.model.train()
 
-    def local_step(i):
-        client.model.train()
+    def inner(x, y):
+        """one local SGD step: fwd + loss + bwd + fused arena step.
+        NO metrics / .item() in the hot loop (those are logging, reference
+        computes them per step only for console output)."""
         client.optimizer.zero_grad()
         with amp(args):
-            loss, _ = inference(client.model, client.criterion,
-                                client.metrics, xs[i % pool_n], ys[i % pool_n])
+            loss = client.criterion(client.model(x), y)
         loss.backward()
         client.optimizer.step(apply_lr=True, apply_in_momentum=True,
                               apply_out_momentum=False)
-        return loss
+
+    use_graph = on_gpu and b.graph != 'off'
+    if use_graph:
+        try:
+            static_x = (xs[0] if isinstance(xs, list) else xs[0]).clone()
+            static_y = ys[0].clone()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    inner(static_x, static_y)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                inner(static_x, static_y)
+        except Exception as e:  # noqa: BLE001
+            if b.graph == 'on':
+                raise
+            print('[bench] hipGraph capture failed (%r), eager path' % e,
+                  flush=True)
+            use_graph = False
+    if use_graph:
+        def local_step(i):
+            static_x.copy_(xs[i % pool_n])
+            static_y.copy_(ys[i % pool_n])
+            graph.replay()
+    else:
+        def local_step(i):
+            inner(xs[i % pool_n], ys[i % pool_n])
 
     def sync():
         args.comm_time.append(0.0)

@@ -25,6 +25,10 @@ def init_distributed(args):
         backend = 'nccl' if torch.cuda.is_available() else 'gloo'
     if backend == 'nccl' and not torch.cuda.is_available():
         backend = 'gloo'
+    if backend == 'nccl':
+        # RCCL for GPU tensors, gloo for the small host-side control
+        # messages (online-client ids, dataset indices, metric averages).
+        backend = 'cpu:gloo,cuda:nccl'
     os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
     os.environ.setdefault('MASTER_PORT', '29500')
     os.environ.setdefault('RANK', '0')

@@ -1,0 +1,134 @@
+# -*- coding: utf-8 -*-
+"""Centered PerFedMe — personalized FL with Moreau envelopes
+(arXiv:2006.08848); parity with reference
+`comms/trainings/federated/centered/perfedme.py:30-166`.
+
+The personal model steps on its own loss with the +lambda*(w_p - w) prox
+term fused into the step kernel; every 5 steps (and at sync) the local copy
+of the global model steps toward the personal model."""
+import time
+
+from fedtorch_amd import ops
+from fedtorch_amd.components.scheduler import adjust_learning_rate
+from fedtorch_amd.components.dataset import load_data_batch
+from fedtorch_amd.trainings.flow import get_current_epoch, is_sync_fed
+from fedtorch_amd.trainings.eval import inference
+from fedtorch_amd.trainings.federated import amp
+from fedtorch_amd.trainings.eval_centered import (
+    do_validate_centered, log_validation_centered)
+from fedtorch_amd.aggregation.centered import (
+    fedavg_aggregation_centered, set_online_clients_centered)
+from fedtorch_amd.logs.logging import (
+    log, logging_sync_time, logging_load_time, logging_globally)
+from fedtorch_amd.logs.meter import define_local_training_tracker
+
+
+def train_and_validate_perfedme_centered(Clients, Server):
+    log('start training and validation of PerFedMe in a centered way.')
+    tracker = define_local_training_tracker()
+    start_global_time = time.time()
+    tracker['start_load_time'] = time.time()
+    args = Server.args
+
+    for n_c in range(args.num_comms):
+        args.rounds_comm += 1
+        args.local_index += 1
+        Server.zero_grad()
+        Server.reset_tracker(Server.local_val_tracker)
+        Server.reset_tracker(Server.global_val_tracker)
+        if args.fed_personal:
+            Server.reset_tracker(Server.local_personal_val_tracker)
+            Server.reset_tracker(Server.global_personal_val_tracker)
+        log('Starting round {} of training'.format(n_c + 1))
+        online_clients = set_online_clients_centered(args)
+
+        for oc in online_clients:
+            C = Clients[oc]
+            C.arena.load_flat(Server.arena.flat)
+            C.args.rounds_comm = args.rounds_comm
+            is_sync = False
+            do_validate_centered(C.args, Server.model, Server.criterion,
+                                 Server.metrics, Server.optimizer,
+                                 C.train_loader, Server.global_val_tracker,
+                                 val=False, local=False)
+            if args.fed_personal:
+                do_validate_centered(C.args, Server.model, Server.criterion,
+                                     Server.metrics, Server.optimizer,
+                                     C.val_loader,
+                                     Server.global_personal_val_tracker,
+                                     val=True, local=False)
+            # prox term lambda*(w_personal - w_local) fused into the
+            # personal step (reference `perfedme.py:99-101`)
+            C.optimizer_personal.set_correction(
+                prox_mu=C.args.perfedme_lambda, server=C.arena.flat)
+            while not is_sync:
+                for _input, _target in C.train_loader:
+                    C.model.train()
+                    C.model_personal.train()
+                    logging_load_time(tracker)
+                    C.args.local_index += 1
+                    C.args.local_data_seen += len(_target)
+                    get_current_epoch(C.args)
+                    lr = adjust_learning_rate(C.args, C.optimizer_personal,
+                                              C.scheduler)
+                    _input, _target = load_data_batch(C.args, _input,
+                                                      _target, tracker)
+                    if _input.size(0) == 1:
+                        is_sync = is_sync_fed(C.args)
+                        break
+                    C.optimizer_personal.zero_grad()
+                    with amp(args):
+                        loss, _ = inference(C.model_personal, C.criterion,
+                                            C.metrics, _input, _target)
+                    loss.backward()
+                    C.optimizer_personal.step(
+                        apply_lr=True, apply_in_momentum=C.args.in_momentum,
+                        apply_out_momentum=False)
+                    is_sync = is_sync_fed(C.args)
+                    if C.args.local_index % 5 == 0 or is_sync:
+                        # local-global update: grad = lambda*(w - w_p)
+                        # (reference `perfedme.py:116-124`)
+                        log('Updating the local version of the global model',
+                            C.args.debug)
+                        lr = adjust_learning_rate(C.args, C.optimizer,
+                                                  C.scheduler)
+                        g = C.work.setdefault('pfm_g',
+                                              C.arena.new_buffer())
+                        ops.scaled_diff(C.arena.flat,
+                                        C.arena_personal.flat, g,
+                                        C.args.perfedme_lambda)
+                        C.optimizer.step(
+                            apply_lr=True,
+                            apply_in_momentum=C.args.in_momentum,
+                            apply_out_momentum=False, grad=g)
+                    tracker['start_load_time'] = time.time()
+                    if is_sync:
+                        break
+            C.optimizer_personal.clear_correction()
+            do_validate_centered(C.args, C.model_personal, C.criterion,
+                                 C.metrics, C.optimizer_personal,
+                                 C.train_loader, Server.local_val_tracker,
+                                 val=False, local=True)
+            if args.fed_personal:
+                do_validate_centered(C.args, C.model_personal, C.criterion,
+                                     C.metrics, C.optimizer_personal,
+                                     C.val_loader,
+                                     Server.local_personal_val_tracker,
+                                     val=True, local=True)
+            tracker['start_sync_time'] = time.time()
+            args.global_index += 1
+            logging_sync_time(tracker)
+
+        fedavg_aggregation_centered(Clients, Server, online_clients)
+        log_validation_centered(args, Server.local_val_tracker, val=False,
+                                local=True)
+        if args.fed_personal:
+            log_validation_centered(args, Server.local_personal_val_tracker,
+                                    val=True, local=True)
+        log_validation_centered(args, Server.global_val_tracker, val=False,
+                                local=False)
+        if args.fed_personal:
+            log_validation_centered(args, Server.global_personal_val_tracker,
+                                    val=True, local=False)
+        logging_globally(tracker, start_global_time)
+        start_global_time = time.time()

@@ -99,9 +99,10 @@ def test_quantize_gpu_matches_eager(bits):
     diff = (q.cpu().long() - qe.long()).abs()
     assert (diff > 1).sum().item() == 0
     assert (diff == 1).float().mean().item() < 1e-3
-    # round-trip error bound holds exactly
+    # round-trip bound: scale/2 quantization + up to 1 grid unit of clipping
+    # from the reference's truncated zero-point (`flow_utils.py:183-192`)
     xr = ops.dequantize(q, info)
-    assert (x - xr).abs().max().item() <= info[0].item() * 0.5 + 1e-6
+    assert (x - xr).abs().max().item() <= info[0].item() * 1.5 + 1e-6
 
 
 def test_dequant_accumulate_gpu():
