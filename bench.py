@@ -44,7 +44,10 @@ def parse():
     p.add_argument('--dtype', type=str, default='bf16')
     p.add_argument('--graph', type=str, default='auto',
                    help='hipGraph-capture the local step: auto|on|off')
-    p.add_argument('--layout', type=str, default='nhwc',
+    # NCHW measured faster than NHWC under hipGraphs on gfx950 (MIOpen falls
+    # back to a naive NHWC wrw kernel for the 3-channel stem; see
+    # profiles/r01_bench_notes.md)
+    p.add_argument('--layout', type=str, default='nchw',
                    choices=['nhwc', 'nchw'])
     return p.parse_args()
 

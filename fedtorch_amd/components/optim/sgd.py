@@ -43,6 +43,17 @@ class FusedSGD(object):
         self._ctrl_client = None
         self._delta = None
 
+    # ---- virtual-client state swap -----------------------------------------
+    def bind_state(self, in_buf=None, out_buf=None, in_init=True,
+                   out_init=True):
+        """Point the momentum buffers at externally owned per-client slices
+        (virtual-client packing keeps [C, N] momentum arenas resident in
+        HBM and swaps views per client)."""
+        self._in_buf = in_buf
+        self._out_buf = out_buf
+        self._in_init = in_init
+        self._out_init = out_init
+
     # ---- correction plumbing ------------------------------------------------
     def set_correction(self, prox_mu=0.0, server=None, ctrl_server=None,
                        ctrl_client=None, delta=None):

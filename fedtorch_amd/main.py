@@ -43,6 +43,16 @@ def main(args):
     client.initialize_dataset()
     client.load_local_dataset()
     client.gen_aux_models()
+    if args.federated and args.clients_per_rank > 1:
+        # virtual-client packing: C clients per GPU rank, replicas resident
+        # in HBM (fedavg / DRFA-over-fedavg).
+        from fedtorch_amd.parallel.multiclient import ClientPack
+        from fedtorch_amd.trainings.packed import \
+            train_and_validate_federated_packed
+        pack = ClientPack(client, args.clients_per_rank)
+        pack.build_loaders()
+        train_and_validate_federated_packed(client, pack)
+        return
     if args.federated:
         if args.federated_drfa:
             from fedtorch_amd.trainings.drfa import \

@@ -126,6 +126,20 @@ class Comm(object):
             self._toc(tok)
         return sorted(t.tolist())
 
+    def sample_online_global(self, total_clients, rate=None):
+        """Sample the online set over W*C VIRTUAL clients (packed mode) and
+        broadcast the ids from rank 0."""
+        import numpy as np
+        rate = rate if rate is not None else self.args.online_client_rate
+        n = max(int(rate * total_clients), 1)
+        onl = np.random.permutation(total_clients)[:n]
+        t = torch.tensor(sorted(onl.tolist()), dtype=torch.int32)
+        if dist_ready():
+            tok = self._tic()
+            dist.broadcast(t, src=0)
+            self._toc(tok)
+        return set(t.tolist())
+
     def gather_scalar(self, value):
         """All ranks learn everyone's scalar (reference `loss_gather`
         `misc.py:54-63` gathers to 0; all-gather keeps every rank able to
