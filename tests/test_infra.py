@@ -1,0 +1,123 @@
+# -*- coding: utf-8 -*-
+"""Infra tests: checkpoint layout + resume, growing-batch sampler, flow
+bookkeeping, tools parsers."""
+import os
+import types
+
+import numpy as np
+import torch
+
+from fedtorch_amd.components.dataset import GrowingMinibatchSampler
+from fedtorch_amd.trainings import flow
+
+
+def test_growing_minibatch_sampler():
+    data = list(range(1000))
+    s = GrowingMinibatchSampler(data, num_epochs=3, base_batch_size=2,
+                                rho=1.05, max_batch_size=64)
+    batches = list(iter(s))
+    sizes = [len(b) for b in batches]
+    assert sizes[0] == 3  # int(2*1.05^0)+1
+    assert all(a <= max(b, 64) for a, b in zip(sizes, sizes[1:] + [64]))
+    assert max(sizes) <= 64
+    # pool can run dry before the nominal iteration count (reference
+    # semantics: batch list precomputed, pool sliced until empty)
+    assert 0 < len(batches) <= s.num_iterations
+    # consuming the sampler twice yields the same batches (idx_pool kept)
+    batches2 = list(iter(s))
+    assert [len(b) for b in batches2] == sizes
+
+
+def _flow_args(**kw):
+    base = dict(growing_batch_size=False, local_index=0, local_data_seen=0,
+                num_batches_train_per_device_per_epoch=10,
+                num_samples_per_epoch=100, stop_criteria='epoch',
+                num_epochs=2, num_iterations_per_worker=50,
+                federated_sync_type='epoch', num_epochs_per_comm=1,
+                local_steps=[4] * 5, epoch=0, epoch_=0.0,
+                client_epoch_total=0)
+    base.update(kw)
+    return types.SimpleNamespace(**base)
+
+
+def test_flow_epoch_and_sync():
+    a = _flow_args(local_index=25)
+    flow.get_current_epoch(a)
+    assert a.epoch_ == 2.5 and a.epoch == 2
+    assert flow.get_current_local_step(a) == 4
+    a.epoch = 99  # falls back to last entry
+    assert flow.get_current_local_step(a) == 4
+    assert flow.is_stop(a)
+
+    a = _flow_args(federated_sync_type='local_step', local_index=8)
+    assert flow.is_sync_fed(a)
+    a.local_index = 7
+    assert not flow.is_sync_fed(a)
+
+
+def test_checkpoint_roundtrip(tmp_path, tiny_args):
+    import torch.nn as nn
+    from fedtorch_amd.logs.checkpoint import (
+        init_checkpoint, save_to_checkpoint, maybe_resume_from_checkpoint)
+    from fedtorch_amd.parallel.arena import Arena
+    from fedtorch_amd.components.optim.sgd import FusedSGD
+
+    args = tiny_args
+    args.checkpoint = str(tmp_path)
+    args.graph = types.SimpleNamespace(rank=0)
+    args.debug = True
+    args.timestamp = 'testrun'
+    init_checkpoint(args)
+    assert args.checkpoint_root.endswith('testrun')
+    assert args.checkpoint_dir.endswith('testrun/0')
+
+    torch.manual_seed(0)
+    model = nn.Linear(5, 2)
+    arena = Arena(model)
+    opt = FusedSGD(arena, lr=0.1, in_momentum=0.9)
+    g = torch.randn_like(arena.grad)
+    opt.step(grad=g)
+    args.best_epoch = [1.0]
+    state = {'arguments': args, 'current_epoch': 1, 'local_index': 7,
+             'global_index': 2, 'arch': 'x',
+             'state_dict': model.state_dict(),
+             'optimizer': opt.state_dict(), 'best_prec1': 55.0}
+    save_to_checkpoint(state, True, dirname=args.checkpoint_root,
+                       filename='checkpoint.pth.tar', save_all=False)
+    assert os.path.exists(os.path.join(args.checkpoint_root,
+                                       'checkpoint.pth.tar'))
+    assert os.path.exists(os.path.join(args.checkpoint_root,
+                                       'model_best.pth.tar'))
+
+    # resume into a fresh model
+    torch.manual_seed(1)
+    model2 = nn.Linear(5, 2)
+    arena2 = Arena(model2)
+    opt2 = FusedSGD(arena2, lr=0.1, in_momentum=0.9)
+    args2 = types.SimpleNamespace(
+        resume=args.checkpoint_root, checkpoint_index=None,
+        data=args.data, batch_size=args.batch_size,
+        num_epochs=args.num_epochs,
+        graph=types.SimpleNamespace(rank=0), best_epoch=[], local_index=0,
+        best_prec1=0)
+    maybe_resume_from_checkpoint(args2, model2, opt2)
+    assert args2.local_index == 7
+    assert args2.best_prec1 == 55.0
+    assert torch.equal(model2.weight.data, model.weight.data)
+    assert torch.allclose(opt2._in_buf, opt._in_buf)
+
+
+def test_tools_parse_roundtrip(tmp_path):
+    from fedtorch_amd.tools.load_console_records import (
+        parse_record_for_test, parse_record_for_train, PAT_COMM)
+    rec = tmp_path / 'record0'
+    rec.write_text(
+        '2026:09:13 10:00:00\tTest at batch: 12. Epoch: 1. Process: 0. '
+        'Prec@1: 55.000 Prec@5: 90.000 Loss: 1.234 Comm: 3\n'
+        '2026:09:13 10:00:05\tTest at batch: 24. Epoch: 2. Process: 0. '
+        'Prec@1: 60.000 Prec@5: 92.000 Loss: 1.100 Comm: 4\n'
+        '2026:09:13 10:00:06\tThis round communication time is: 0.125\n')
+    df = parse_record_for_test(str(rec))
+    assert len(df) == 2
+    assert df['top1'].tolist() == [55.0, 60.0]
+    assert df['time'].tolist() == [0.0, 5.0]
