@@ -226,6 +226,17 @@ def scatter_accumulate(out, vs, idxs):
         out.scatter_add_(0, idxs[kk].long(), vs[kk])
 
 
+def multi_diff_accumulate(server, replicas, weights, out):
+    """out = sum_c weights[c] * (server - replicas[c]) — ONE pass over the
+    [C, N] replica arena (packed virtual clients)."""
+    wsum = float(weights.sum())
+    if _use_hip(server, replicas, out):
+        return _C.multi_diff_accumulate(server, replicas, weights, out,
+                                        wsum)
+    torch.sum((server.unsqueeze(0) - replicas) * weights.view(-1, 1), dim=0,
+              out=out)
+
+
 def error_feedback_update(mem, grad, d, inv_weight):
     """mem += grad * inv_weight - d (reference `fedgate.py:81`,
     `qsparse.py:57`: `memory += grad/rank_weight - d`)."""

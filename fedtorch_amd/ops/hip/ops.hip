@@ -630,6 +630,51 @@ std::vector<torch::Tensor> topk_compress(torch::Tensor x, long k) {
   return {v, idx};
 }
 
+// batched virtual-client aggregation: ONE pass over the [C, N] replica
+// arena — out[i] = sum_c w[c] * (server[i] - replicas[c][i]).
+// (packed mode, fedtorch_amd/parallel/multiclient.py; replaces C separate
+// diff+add launches.)
+__global__ void multi_diff_acc_kernel(const float* __restrict__ server,
+                                      const float* __restrict__ replicas,
+                                      const float* __restrict__ w, long C,
+                                      long n4, float wsum,
+                                      float* __restrict__ out) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 sv = reinterpret_cast<const float4*>(server)[i];
+    const float* ss = reinterpret_cast<const float*>(&sv);
+    float acc[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[j] = wsum * ss[j];
+    for (long c = 0; c < C; ++c) {
+      float wc = w[c];
+      if (wc == 0.f) continue;
+      float4 rv = reinterpret_cast<const float4*>(replicas + c * n4 * 4)[i];
+      const float* rr = reinterpret_cast<const float*>(&rv);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] = fmaf(-wc, rr[j], acc[j]);
+    }
+    float4 ov;
+    float* oo = reinterpret_cast<float*>(&ov);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) oo[j] = acc[j];
+    reinterpret_cast<float4*>(out)[i] = ov;
+  }
+}
+
+void multi_diff_accumulate(torch::Tensor server, torch::Tensor replicas,
+                           torch::Tensor weights, torch::Tensor out,
+                           double wsum) {
+  CHK(server); CHK(replicas); CHK(out);
+  long C = replicas.size(0);
+  long n4 = server.numel() / 4;
+  hipLaunchKernelGGL(multi_diff_acc_kernel, dim3(ft_grid(n4)),
+                     dim3(FT_BLOCK), 0, STREAM, server.data_ptr<float>(),
+                     replicas.data_ptr<float>(), weights.data_ptr<float>(),
+                     C, n4, (float)wsum, out.data_ptr<float>());
+}
+
 // fused decompress + K-way sum: out = sum_k scatter(vs[k] @ idxs[k]) -------
 __global__ void scatter_acc_kernel(const float* __restrict__ vs,
                                    const int* __restrict__ idxs, long total,
@@ -665,4 +710,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequant_accumulate", &dequant_accumulate);
   m.def("topk_compress", &topk_compress);
   m.def("scatter_accumulate", &scatter_accumulate);
+  m.def("multi_diff_accumulate", &multi_diff_accumulate);
 }

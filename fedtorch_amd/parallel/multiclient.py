@@ -101,13 +101,11 @@ class ClientPack(object):
         return steps
 
     def accumulate_partial(self, server_flat, weights):
-        """partial = sum_j weights[j] * (server - replica_j); weights: list
-        of per-local-client floats (0 for offline)."""
-        self.partial.zero_()
-        diff = torch.empty_like(self.partial)
-        for j, w in enumerate(weights):
-            if w == 0.0:
-                continue
-            ops.scaled_diff(server_flat, self.replicas[j], diff, w)
-            self.partial.add_(diff)
+        """partial = sum_j weights[j] * (server - replica_j) — ONE batched
+        kernel over the [C, N] replica arena; weights: list of per-local-
+        client floats (0 for offline)."""
+        w = torch.tensor(weights, dtype=torch.float32,
+                         device=server_flat.device)
+        ops.multi_diff_accumulate(server_flat, self.replicas, w,
+                                  self.partial)
         return self.partial

@@ -200,3 +200,16 @@ def test_scaffold_delta_gpu():
     agg = torch.randn(4096, device='cuda')
     ops.delta_update(delta, s, agg, c, 0.25)
     assert torch.allclose(delta, d0 + 0.25 * (s - agg - c), atol=1e-5)
+
+
+def test_multi_diff_accumulate_gpu():
+    torch.manual_seed(13)
+    C, n = 7, 8192
+    server = torch.randn(n, device='cuda')
+    replicas = torch.randn(C, n, device='cuda')
+    w = torch.rand(C, device='cuda')
+    w[2] = 0.0
+    out = torch.empty(n, device='cuda')
+    ops.multi_diff_accumulate(server, replicas, w, out)
+    expected = ((server.unsqueeze(0) - replicas) * w.view(-1, 1)).sum(0)
+    assert torch.allclose(out, expected, atol=1e-4)
