@@ -33,4 +33,4 @@ main(get_args(['-d','cifar10','-a','resnet20','-f','true','--federated_type','sc
  '--num_comms','2','--online_client_rate','0.5','--num_workers','8','-b','64','--lr','0.1',
  '--bf16','true','--checkpoint','/tmp/ckcen','--debug','false']))
 print('CENTERED-GPU OK')" > gpurun_out/centered_gpu.log 2>&1; echo CENTERED=$?
-tail -2 gpurun_out/quant_gpu.log gpurun_out/comp_gpu.log gpurun_out/packed_gpu.log gpurun_out/centered_gpu.log
+for f in gpurun_out/quant_gpu.log gpurun_out/comp_gpu.log gpurun_out/packed_gpu.log gpurun_out/centered_gpu.log; do tail -1 $f; done
