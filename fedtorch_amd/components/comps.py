@@ -25,6 +25,13 @@ def create_components(args):
         model = model.cuda()
         if getattr(args, 'channels_last', False):
             model = model.to(memory_format=torch.channels_last)
+        if getattr(args, 'fused_bn', True) and \
+                getattr(args, 'hip_kernels', True) and \
+                not getattr(args, 'channels_last', False):
+            from fedtorch_amd.ops.batchnorm import convert_to_fused_bn
+            import fedtorch_amd.ops as _ops
+            if _ops.hip_available():
+                convert_to_fused_bn(model)
     arena = Arena(model)
     criterion = define_criterion(args)
     if args.graph.on_cuda and torch.cuda.is_available():

@@ -12,6 +12,7 @@
 #include <hip/hip_runtime.h>
 
 #include "common.h"
+#include "batchnorm.h"
 
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 #define STREAM at::hip::getCurrentHIPStream().stream()
@@ -695,7 +696,93 @@ void scatter_accumulate(torch::Tensor out, torch::Tensor vs,
                      total, out.data_ptr<float>());
 }
 
+// ==========================================================================
+// fused spatial BatchNorm (training fwd/bwd) — kernels in batchnorm.h
+// ==========================================================================
+static inline int bn_chunks(long per_ch, long C) {
+  long by_work = (per_ch + FT_BLOCK * 8 - 1) / (FT_BLOCK * 8);
+  long cap = FT_MAX_BLOCKS / (C > 0 ? C : 1);
+  long c = by_work < cap ? by_work : cap;
+  return (int)(c < 1 ? 1 : c);
+}
+
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
+                                        torch::Tensor bias,
+                                        torch::Tensor running_mean,
+                                        torch::Tensor running_var,
+                                        double eps, double momentum,
+                                        bool relu) {
+  CHK(x);
+  TORCH_CHECK(x.dim() == 4, "bn_fwd_train expects NCHW");
+  long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto stats = torch::zeros({C, 2}, f32);
+  auto save_mean = torch::empty({C}, f32);
+  auto save_ivar = torch::empty({C}, f32);
+  auto y = torch::empty_like(x);
+  dim3 grid(bn_chunks(N * HW, C), C);
+  bool track = running_mean.numel() > 0;
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "bn_fwd", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                 __hip_bfloat16, scalar_t>;
+    hipLaunchKernelGGL(bn_stats_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+                       reinterpret_cast<const T*>(x.data_ptr()), N, C, HW,
+                       stats.data_ptr<float>());
+    hipLaunchKernelGGL(bn_norm_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       reinterpret_cast<T*>(y.data_ptr()),
+                       stats.data_ptr<float>(),
+                       weight.numel() ? weight.data_ptr<float>() : nullptr,
+                       bias.numel() ? bias.data_ptr<float>() : nullptr,
+                       save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(),
+                       track ? running_mean.data_ptr<float>() : nullptr,
+                       track ? running_var.data_ptr<float>() : nullptr,
+                       N, C, HW, (float)eps, (float)momentum, relu ? 1 : 0);
+  });
+  return {y, save_mean, save_ivar};
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor y, torch::Tensor save_mean,
+                                  torch::Tensor save_ivar,
+                                  torch::Tensor weight, bool relu) {
+  CHK(dy); CHK(x);
+  long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto red = torch::zeros({C, 2}, f32);
+  auto dx = torch::empty_like(x);
+  auto dweight = torch::empty({C}, f32);
+  auto dbias = torch::empty({C}, f32);
+  dim3 grid(bn_chunks(N * HW, C), C);
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "bn_bwd", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                 __hip_bfloat16, scalar_t>;
+    const T* yp = relu ? reinterpret_cast<const T*>(y.data_ptr()) : nullptr;
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<T>, grid, dim3(FT_BLOCK), 0,
+                       STREAM, reinterpret_cast<const T*>(dy.data_ptr()),
+                       reinterpret_cast<const T*>(x.data_ptr()), yp,
+                       save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(), N, C, HW,
+                       red.data_ptr<float>(), relu ? 1 : 0);
+    hipLaunchKernelGGL(bn_bwd_dx_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+                       reinterpret_cast<const T*>(dy.data_ptr()),
+                       reinterpret_cast<const T*>(x.data_ptr()), yp,
+                       red.data_ptr<float>(), save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(),
+                       weight.numel() ? weight.data_ptr<float>() : nullptr,
+                       reinterpret_cast<T*>(dx.data_ptr()),
+                       dweight.data_ptr<float>(), dbias.data_ptr<float>(),
+                       N, C, HW, relu ? 1 : 0);
+  });
+  return {dx, dweight, dbias};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_bwd", &bn_bwd);
   m.def("fused_sgd_step", &fused_sgd_step, "fused dual-mode SGD step");
   m.def("weighted_diff_restore", &weighted_diff_restore);
   m.def("scaled_diff", &scaled_diff);

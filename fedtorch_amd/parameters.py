@@ -205,6 +205,9 @@ def build_parser():
                         help='virtual clients packed per GPU rank (their '
                              'replicas and aux state stay resident in HBM3E).')
     parser.add_argument('--channels_last', type=str2bool, default=False)
+    parser.add_argument('--fused_bn', type=str2bool, default=True,
+                        help='replace nn.BatchNorm2d with the fused gfx950 '
+                             'BN kernels for GPU training.')
     return parser
 
 

@@ -1,0 +1,43 @@
+#!/usr/bin/env python3
+"""Summarize a rocprofv3 sqlite results db into a small text report
+(top kernels by total time) — run ON the GPU box so only the summary ships
+back through gpurun_out."""
+import glob
+import sqlite3
+import sys
+
+
+def summarize(db_path, out_path, top=30):
+    db = sqlite3.connect(db_path)
+    cur = db.cursor()
+    tables = [r[0] for r in cur.execute(
+        "SELECT name FROM sqlite_master WHERE type='table' "
+        "AND name LIKE 'rocpd_kernel_dispatch%'")]
+    if not tables:
+        raise SystemExit('no kernel dispatch table in %s' % db_path)
+    sfx = tables[0].replace('rocpd_kernel_dispatch_', '')
+    tot = cur.execute(
+        "SELECT SUM(end-start)/1e6, COUNT(*), (MAX(end)-MIN(start))/1e6 "
+        "FROM rocpd_kernel_dispatch_%s" % sfx).fetchone()
+    lines = ['db: %s' % db_path,
+             'TOTAL kernel %.1f ms / %d dispatches; wall span %.1f ms' % tot]
+    q = ("SELECT ks.display_name, COUNT(*), SUM(k.end-k.start)/1e6, "
+         "AVG(k.end-k.start)/1e3 FROM rocpd_kernel_dispatch_%s k "
+         "JOIN rocpd_info_kernel_symbol_%s ks ON k.kernel_id=ks.id "
+         "GROUP BY ks.display_name ORDER BY SUM(k.end-k.start) DESC "
+         "LIMIT %d" % (sfx, sfx, top))
+    for name, calls, tot_ms, avg_us in cur.execute(q):
+        lines.append('%9.2f ms %7dx %8.1fus  %s'
+                     % (tot_ms, calls, avg_us, name[:100]))
+    with open(out_path, 'w') as f:
+        f.write('\n'.join(lines) + '\n')
+    print('\n'.join(lines[:12]))
+
+
+if __name__ == '__main__':
+    pattern = sys.argv[1] if len(sys.argv) > 1 else 'gpurun_out/prof*/**/*.db'
+    out = sys.argv[2] if len(sys.argv) > 2 else 'gpurun_out/prof_summary.txt'
+    dbs = sorted(glob.glob(pattern, recursive=True))
+    if not dbs:
+        raise SystemExit('no dbs matching %s' % pattern)
+    summarize(dbs[-1], out)

@@ -213,3 +213,52 @@ def test_multi_diff_accumulate_gpu():
     ops.multi_diff_accumulate(server, replicas, w, out)
     expected = ((server.unsqueeze(0) - replicas) * w.view(-1, 1)).sum(0)
     assert torch.allclose(out, expected, atol=1e-4)
+
+
+def test_fused_batchnorm_matches_torch():
+    """fused BN fwd/bwd vs nn.BatchNorm2d on identical fp32 inputs."""
+    import torch.nn as nn
+    from fedtorch_amd.ops.batchnorm import FusedBatchNorm2d
+    torch.manual_seed(20)
+    N, C, H, W = 16, 32, 14, 14
+    x = torch.randn(N, C, H, W, device='cuda')
+    ref_bn = nn.BatchNorm2d(C).cuda()
+    fus_bn = FusedBatchNorm2d(C).cuda()
+    fus_bn.load_state_dict(ref_bn.state_dict())
+    ref_bn.weight.data.uniform_(0.5, 1.5)
+    ref_bn.bias.data.uniform_(-0.5, 0.5)
+    fus_bn.weight.data.copy_(ref_bn.weight.data)
+    fus_bn.bias.data.copy_(ref_bn.bias.data)
+
+    xr = x.clone().requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    yr = ref_bn(xr)
+    yf = fus_bn(xf)
+    assert torch.allclose(yf, yr, atol=2e-5), \
+        (yf - yr).abs().max().item()
+    g = torch.randn_like(yr)
+    yr.backward(g)
+    yf.backward(g)
+    assert torch.allclose(xf.grad, xr.grad, atol=2e-4)
+    assert torch.allclose(fus_bn.weight.grad, ref_bn.weight.grad, atol=1e-3)
+    assert torch.allclose(fus_bn.bias.grad, ref_bn.bias.grad, atol=1e-3)
+    assert torch.allclose(fus_bn.running_mean, ref_bn.running_mean,
+                          atol=1e-5)
+    assert torch.allclose(fus_bn.running_var, ref_bn.running_var, atol=1e-4)
+
+
+def test_fused_batchnorm_bf16_io():
+    from fedtorch_amd.ops.batchnorm import FusedBatchNorm2d
+    torch.manual_seed(21)
+    x = torch.randn(8, 16, 8, 8, device='cuda', dtype=torch.bfloat16)
+    bn = FusedBatchNorm2d(16).cuda()
+    xr = x.clone().requires_grad_(True)
+    y = bn(xr)
+    assert y.dtype == torch.bfloat16
+    y.float().square().mean().backward()
+    assert xr.grad is not None and torch.isfinite(xr.grad.float()).all()
+    # fp32 reference on the same values
+    import torch.nn as nn
+    ref = nn.BatchNorm2d(16).cuda()
+    yr = ref(x.float())
+    assert torch.allclose(y.float(), yr, atol=0.05)
