@@ -49,6 +49,8 @@ def parse():
     # profiles/r01_bench_notes.md)
     p.add_argument('--layout', type=str, default='nchw',
                    choices=['nhwc', 'nchw'])
+    p.add_argument('--fused_bn', type=str, default='on',
+                   choices=['on', 'off'])
     return p.parse_args()
 
 
@@ -74,6 +76,7 @@ def main():
         '--bf16', 'true' if use_bf16 else 'false',
         '--channels_last',
         'true' if (on_gpu and b.layout == 'nhwc') else 'false',
+        '--fused_bn', 'true' if b.fused_bn == 'on' else 'false',
         '-j', '0', '--checkpoint', '/tmp/ft_bench_ckpt', '--debug', 'false'])
     os.environ.setdefault('FEDTORCH_SYNTH_SIZE', '2048')
 

@@ -187,7 +187,23 @@ def train_and_validate_federated(client, validate=True):
         log('Enter synching', args.debug)
         tracker['start_sync_time'] = time.time()
         args.global_index += 1
+        if args.check_model_at_sync:
+            from fedtorch_amd.logs.check_training import check_model_at_sync
+            check_model_at_sync(args, client.arena, tag='pre-sync')
+        if args.track_model_aggregation:
+            local_flat_before = client.arena.clone_flat()
+            server_before = client.model_server.clone()
         aggregate_round(client, online_clients, lr, local_steps)
+        if args.track_model_aggregation:
+            from fedtorch_amd.logs.check_training import \
+                track_model_aggregation
+            track_model_aggregation(
+                args,
+                local_diff_flat=local_flat_before - server_before,
+                agg_diff_flat=client.arena.flat - server_before,
+                init_flat=client.work.setdefault(
+                    'init_flat', server_before.clone()),
+                current_flat=client.arena.flat)
         client.comm.flush_comm_time()
         logging_sync_time(tracker)
 
