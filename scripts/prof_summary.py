@@ -7,7 +7,9 @@ import sqlite3
 import sys
 
 
-def summarize(db_path, out_path, top=30):
+def summarize(db_path, out_path, top=30, window=0.0):
+    """window: fraction of the timeline to SKIP from the start (e.g. 0.5
+    keeps only the second half — steady state, past MIOpen find/warmup)."""
     db = sqlite3.connect(db_path)
     cur = db.cursor()
     tables = [r[0] for r in cur.execute(
@@ -16,16 +18,22 @@ def summarize(db_path, out_path, top=30):
     if not tables:
         raise SystemExit('no kernel dispatch table in %s' % db_path)
     sfx = tables[0].replace('rocpd_kernel_dispatch_', '')
+    t0, t1 = cur.execute(
+        "SELECT MIN(start), MAX(end) FROM rocpd_kernel_dispatch_%s"
+        % sfx).fetchone()
+    cut = t0 + (t1 - t0) * window
+    where = "WHERE k.start >= %d" % cut
     tot = cur.execute(
-        "SELECT SUM(end-start)/1e6, COUNT(*), (MAX(end)-MIN(start))/1e6 "
-        "FROM rocpd_kernel_dispatch_%s" % sfx).fetchone()
-    lines = ['db: %s' % db_path,
+        "SELECT SUM(k.end-k.start)/1e6, COUNT(*), "
+        "(MAX(k.end)-MIN(k.start))/1e6 "
+        "FROM rocpd_kernel_dispatch_%s k %s" % (sfx, where)).fetchone()
+    lines = ['db: %s (window skip=%.0f%%)' % (db_path, window * 100),
              'TOTAL kernel %.1f ms / %d dispatches; wall span %.1f ms' % tot]
     q = ("SELECT ks.display_name, COUNT(*), SUM(k.end-k.start)/1e6, "
          "AVG(k.end-k.start)/1e3 FROM rocpd_kernel_dispatch_%s k "
-         "JOIN rocpd_info_kernel_symbol_%s ks ON k.kernel_id=ks.id "
+         "JOIN rocpd_info_kernel_symbol_%s ks ON k.kernel_id=ks.id %s "
          "GROUP BY ks.display_name ORDER BY SUM(k.end-k.start) DESC "
-         "LIMIT %d" % (sfx, sfx, top))
+         "LIMIT %d" % (sfx, sfx, where, top))
     for name, calls, tot_ms, avg_us in cur.execute(q):
         lines.append('%9.2f ms %7dx %8.1fus  %s'
                      % (tot_ms, calls, avg_us, name[:100]))
@@ -37,7 +45,8 @@ def summarize(db_path, out_path, top=30):
 if __name__ == '__main__':
     pattern = sys.argv[1] if len(sys.argv) > 1 else 'gpurun_out/prof*/**/*.db'
     out = sys.argv[2] if len(sys.argv) > 2 else 'gpurun_out/prof_summary.txt'
+    window = float(sys.argv[3]) if len(sys.argv) > 3 else 0.0
     dbs = sorted(glob.glob(pattern, recursive=True))
     if not dbs:
         raise SystemExit('no dbs matching %s' % pattern)
-    summarize(dbs[-1], out)
+    summarize(dbs[-1], out, window=window)
