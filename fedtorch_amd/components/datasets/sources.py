@@ -150,9 +150,15 @@ def get_vision_dataset(name, root, split, seed=1234):
         log('WARNING: no on-disk files for dataset {} under {} — generating '
             'a deterministic SYNTHETIC stand-in ({} samples, seed {}).'
             .format(name, root, n, seed), debug=True)
+        # LEARNABLE synthetic stand-in: class-dependent Gaussian means, so
+        # convergence / rounds-to-accuracy stay meaningful without real
+        # data. Train and test share templates (same base seed).
+        gt = torch.Generator().manual_seed(seed * 1000 + 7)
+        templates = torch.randn((num_classes,) + shape, generator=gt) * 0.4
         g = torch.Generator().manual_seed(seed + (0 if train else 1))
-        x = torch.rand((n,) + shape, generator=g)
         y = torch.randint(0, num_classes, (n,), generator=g)
+        x = (torch.rand((n,) + shape, generator=g) * 0.8 +
+             templates[y]).clamp_(0, 1)
     else:
         x, y = data
     # normalize
