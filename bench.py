@@ -51,6 +51,8 @@ def parse():
                    choices=['nhwc', 'nchw'])
     p.add_argument('--fused_bn', type=str, default='on',
                    choices=['on', 'off'])
+    p.add_argument('--fp32_stem', type=str, default='off',
+                   choices=['on', 'off'])
     return p.parse_args()
 
 
@@ -89,6 +91,9 @@ def main():
     client = Client(args, rank)
     client.initialize()
     client.gen_aux_models()
+    if b.fp32_stem == 'on' and on_gpu:
+        from fedtorch_amd.components.models.resnet import _Fp32Stem
+        client.model.conv1 = _Fp32Stem(client.model.conv1)
     args = client.args
     # pretend dataset-derived counters (synthetic pool below replaces loaders)
     args.num_batches_train_per_device_per_epoch = 390

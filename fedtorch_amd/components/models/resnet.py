@@ -9,6 +9,7 @@
 Written channels-last friendly (contiguous module graph, no functional
 padding games) so MIOpen picks NHWC kernels under bf16 autocast on gfx950.
 """
+import torch
 import torch.nn as nn
 
 _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10, 'mnist': 10,
@@ -95,6 +96,20 @@ class _ResNetBase(nn.Module):
             elif isinstance(m, nn.BatchNorm2d):
                 nn.init.constant_(m.weight, 1)
                 nn.init.constant_(m.bias, 0)
+
+
+class _Fp32Stem(nn.Module):
+    """Run the 3-channel stem conv in fp32 even under bf16 autocast: MIOpen
+    has no tuned bf16 NHWC wrw kernel for 3-input-channel convs and falls
+    back to a naive kernel (profiles/r01_bench_notes.md)."""
+
+    def __init__(self, conv):
+        super().__init__()
+        self.conv = conv
+
+    def forward(self, x):
+        with torch.autocast('cuda', enabled=False):
+            return self.conv(x.float())
 
 
 class ResNetCifar(_ResNetBase):
