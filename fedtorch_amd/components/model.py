@@ -8,12 +8,13 @@ from fedtorch_amd.logs.logging import log
 
 
 def define_model(args):
-    if 'resnet' in args.arch:
-        model = models.resnet(args)
+    # order matters: 'wideresnet' contains 'resnet'
+    if 'wideresnet' in args.arch:
+        model = models.wideresnet(args)
     elif 'densenet' in args.arch:
         model = models.densenet(args)
-    elif 'wideresnet' in args.arch:
-        model = models.wideresnet(args)
+    elif 'resnet' in args.arch:
+        model = models.resnet(args)
     elif args.arch in ('mlp', 'robust_mlp', 'cnn', 'rnn',
                        'logistic_regression', 'robust_logistic_regression',
                        'least_square', 'robust_least_square'):
