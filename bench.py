@@ -29,7 +29,8 @@ from fedtorch_amd.parameters import get_args  # noqa: E402
 from fedtorch_amd.nodes import Client  # noqa: E402
 from fedtorch_amd.trainings.federated import amp  # noqa: E402
 from fedtorch_amd.trainings.eval import inference  # noqa: E402
-from fedtorch_amd.aggregation.federated import fedavg_aggregation  # noqa: E402
+from fedtorch_amd.aggregation.federated import (  # noqa: E402
+    fedavg_aggregation, aggregate_bn_buffers)
 
 TAU = 10  # local steps per communication round (BASELINE config 2)
 
@@ -163,6 +164,8 @@ def main():
         fedavg_aggregation(args, client.comm, client.arena,
                            client.model_server, client.optimizer, online,
                            work=client.work)
+        aggregate_bn_buffers(args, client.comm, client.arena, online,
+                             work=client.work)
 
     def run(n_steps):
         for s in range(1, n_steps + 1):

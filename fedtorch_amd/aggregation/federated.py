@@ -55,6 +55,32 @@ def _buf(work, name, like):
     return work[name]
 
 
+def aggregate_bn_buffers(args, comm, arena, online_clients, work=None):
+    """Average BatchNorm running stats over the ONLINE clients (uniform
+    mean; offline ranks contribute zero and adopt the mean).  The reference
+    never syncs buffers — see `parallel/arena.py` docstring."""
+    if arena.buf_flat is None or \
+            not getattr(args, 'aggregate_bn_stats', True):
+        return
+    work = work if work is not None else {}
+    w = 1.0 / len(online_clients) \
+        if args.graph.rank in online_clients else 0.0
+    tmp = _buf(work, 'bn_stats_buf', arena.buf_flat)
+    tmp.copy_(arena.buf_flat).mul_(w)
+    comm.all_reduce(tmp)
+    arena.buf_flat.copy_(tmp)
+
+
+def aggregate_bn_buffers_centered(Clients, Server, online_clients):
+    if Server.arena.buf_flat is None or \
+            not getattr(Server.args, 'aggregate_bn_stats', True):
+        return
+    Server.arena.buf_flat.zero_()
+    w = 1.0 / len(online_clients)
+    for o in online_clients:
+        Server.arena.buf_flat.add_(Clients[o].arena.buf_flat, alpha=w)
+
+
 def distribute_model_server(comm, server_flat, src=0):
     """ONE arena broadcast (reference loops P params,
     `federated/misc.py:22-27`)."""

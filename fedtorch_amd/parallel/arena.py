@@ -77,6 +77,30 @@ class Arena(object):
         if with_grads:
             self.grad = torch.zeros_like(self.flat)
             self.attach_grads()
+        # BatchNorm running stats live in a secondary flat buffer so sync
+        # can average them (the reference never aggregates buffers — a
+        # BN-model server would keep INIT stats and its eval-mode outputs
+        # explode; its centered experiments only used BN-free models or
+        # track_running_stats=False, `models/nonconvex/mlp.py:25`).
+        self.buf_flat = None
+        self._pack_running_stats(module)
+
+    def _pack_running_stats(self, module):
+        entries = []
+        total = 0
+        for m in module.modules():
+            for bname in ('running_mean', 'running_var'):
+                b = getattr(m, '_buffers', {}).get(bname, None)
+                if b is not None and b.is_floating_point():
+                    entries.append((m, bname, total, b.numel()))
+                    total += _aligned(b.numel())
+        if not entries:
+            return
+        self.buf_flat = torch.zeros(total, dtype=self.dtype,
+                                    device=self.device)
+        for m, bname, off, n in entries:
+            self.buf_flat[off:off + n].copy_(m._buffers[bname].reshape(-1))
+            m._buffers[bname] = self.buf_flat[off:off + n]
 
     # ---- gradient plumbing -------------------------------------------------
     def attach_grads(self):

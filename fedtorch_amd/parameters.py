@@ -205,6 +205,11 @@ def build_parser():
                         help='virtual clients packed per GPU rank (their '
                              'replicas and aux state stay resident in HBM3E).')
     parser.add_argument('--channels_last', type=str2bool, default=False)
+    parser.add_argument('--aggregate_bn_stats', type=str2bool, default=True,
+                        help='average BatchNorm running stats across online '
+                             'clients at sync (the reference never syncs '
+                             'buffers, leaving the server model with init '
+                             'stats).')
     parser.add_argument('--fused_bn', type=str2bool, default=True,
                         help='replace nn.BatchNorm2d with the fused gfx950 '
                              'BN kernels for GPU training.')

@@ -127,6 +127,8 @@ def train_and_validate_federated_centered(Clients, Server):
         local_steps, lr = 0, args.old_learning_rate
         for oc in online_clients:
             Clients[oc].arena.load_flat(Server.arena.flat)
+            if Server.arena.buf_flat is not None:
+                Clients[oc].arena.buf_flat.copy_(Server.arena.buf_flat)
             Clients[oc].args.rounds_comm = args.rounds_comm
 
             if args.federated_type == 'qffl':
@@ -197,6 +199,9 @@ def train_and_validate_federated_centered(Clients, Server):
             qffl_aggregation_centered(Clients, Server, online_clients, lr)
         else:
             fedavg_aggregation_centered(Clients, Server, online_clients)
+        from fedtorch_amd.aggregation.federated import \
+            aggregate_bn_buffers_centered
+        aggregate_bn_buffers_centered(Clients, Server, online_clients)
 
         log_validation_centered(args, Server.local_val_tracker, val=False,
                                 local=True)

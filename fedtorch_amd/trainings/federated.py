@@ -127,6 +127,9 @@ def aggregate_round(client, online_clients, lr, local_steps,
             args, client.comm, client.arena, client.model_server,
             client.optimizer, online_clients, lambda_weight=lambda_weight,
             work=client.work)
+    from fedtorch_amd.aggregation.federated import aggregate_bn_buffers
+    aggregate_bn_buffers(args, client.comm, client.arena, online_clients,
+                         work=client.work)
 
 
 def train_and_validate_federated(client, validate=True):

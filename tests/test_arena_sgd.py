@@ -144,3 +144,24 @@ def test_sync_step_with_out_momentum():
              apply_out_momentum=True, grad=agg)
     # second: buf = 0.8*agg + agg; p -= 0.5*buf
     assert torch.allclose(arena.flat, flat1 - 0.5 * (1.8 * agg), atol=1e-5)
+
+
+def test_arena_packs_bn_running_stats():
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    m = nn.Sequential(nn.Conv2d(3, 4, 3), nn.BatchNorm2d(4), nn.ReLU())
+    arena = Arena(m)
+    assert arena.buf_flat is not None
+    bn = m[1]
+    # views point into the buffer arena
+    base = arena.buf_flat.data_ptr()
+    end = base + arena.buf_flat.numel() * 4
+    assert base <= bn.running_mean.data_ptr() < end
+    assert base <= bn.running_var.data_ptr() < end
+    # a forward updates stats inside the flat buffer
+    m.train()
+    m(torch.randn(8, 3, 8, 8))
+    assert arena.buf_flat.abs().sum() > 0
+    # no-BN model: no buffer arena
+    arena2 = Arena(nn.Linear(4, 2))
+    assert arena2.buf_flat is None

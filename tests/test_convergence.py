@@ -45,3 +45,29 @@ def test_compressed_fedgate_learns():
                                             '--compressed_ratio', '0.4'])
     acc = server.global_test_tracker['top1'].avg
     assert acc > 30.0, 'compressed top1 %.1f' % acc
+
+
+def test_centered_resnet_bn_stats_aggregated():
+    """ResNet-20 federated training must LEARN (the reference never syncs
+    BN running stats, so a BN-model server diverges in eval; we aggregate
+    them — `aggregation/federated.py:aggregate_bn_buffers`)."""
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '512'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes.centered import ClientCentered, ServerCentered
+    from fedtorch_amd.trainings.centered.main import (
+        train_and_validate_federated_centered)
+    args = get_args(['-d', 'cifar10', '-a', 'resnet20', '-f', 'true',
+                     '--federated_type', 'fedavg', '--num_comms', '6',
+                     '--online_client_rate', '1.0', '-b', '32', '--lr',
+                     '0.05', '--in_momentum', 'true', '--on_cuda', 'false',
+                     '--checkpoint', '/tmp/ft_conv_rn', '--debug', 'false',
+                     '--manual_seed', '5'])
+    args.num_workers = 2
+    Clients = {0: ClientCentered(args, 0)}
+    Clients[1] = ClientCentered(args, 1, Partitioner=Clients[0].Partitioner)
+    Server = ServerCentered(Clients[0].args, Clients[0].model)
+    train_and_validate_federated_centered(Clients, Server)
+    acc = Server.global_test_tracker['top1'].avg
+    loss = Server.global_test_tracker['losses'].avg
+    assert loss < 2.5, 'server eval loss %.2f (BN stats broken?)' % loss
+    assert acc > 25.0, 'test top1 %.1f' % acc
