@@ -30,7 +30,7 @@ from fedtorch_amd.nodes import Client  # noqa: E402
 from fedtorch_amd.trainings.federated import amp  # noqa: E402
 from fedtorch_amd.trainings.eval import inference  # noqa: E402
 from fedtorch_amd.aggregation.federated import (  # noqa: E402
-    fedavg_aggregation, aggregate_bn_buffers)
+    fedavg_aggregation, fedgate_aggregation, aggregate_bn_buffers)
 
 TAU = 10  # local steps per communication round (BASELINE config 2)
 
@@ -54,6 +54,11 @@ def parse():
                    choices=['on', 'off'])
     p.add_argument('--fp32_stem', type=str, default='off',
                    choices=['on', 'off'])
+    p.add_argument('--algo', type=str, default='fedavg',
+                   choices=['fedavg', 'comgate_topk', 'comgate_quant'],
+                   help='sync algorithm: plain FedAvg or FedCOMGATE '
+                        '(FedGATE + top-k compression / quantization, '
+                        'BASELINE config 3)')
     return p.parse_args()
 
 
@@ -70,9 +75,15 @@ def main():
         world = dist.get_world_size()
 
     use_bf16 = b.dtype == 'bf16' and on_gpu
-    args = get_args([
+    fed_type = 'fedgate' if b.algo.startswith('comgate') else 'fedavg'
+    argv_extra = []
+    if b.algo == 'comgate_topk':
+        argv_extra = ['--compressed', 'true', '--compressed_ratio', '0.2']
+    elif b.algo == 'comgate_quant':
+        argv_extra = ['--quantized', 'true', '--quantized_bits', '8']
+    args = get_args(argv_extra + [
         '-d', 'cifar10', '-a', b.model, '-f', 'true',
-        '--federated_type', 'fedavg', '--num_comms', '1000000',
+        '--federated_type', fed_type, '--num_comms', '1000000',
         '--online_client_rate', '1.0', '--federated_sync_type', 'local_step',
         '--local_step', str(TAU), '-b', str(b.batch), '--lr', '0.1',
         '--in_momentum', 'true', '--weight_decay', '5e-4',
@@ -161,9 +172,15 @@ def main():
 
     def sync():
         args.comm_time.append(0.0)
-        fedavg_aggregation(args, client.comm, client.arena,
-                           client.model_server, client.optimizer, online,
-                           work=client.work)
+        if b.algo.startswith('comgate'):
+            fedgate_aggregation(args, client.comm, client.arena,
+                                client.model_server, client.model_delta,
+                                client.model_memory, client.optimizer,
+                                online, lr, TAU, work=client.work)
+        else:
+            fedavg_aggregation(args, client.comm, client.arena,
+                               client.model_server, client.optimizer,
+                               online, work=client.work)
         aggregate_bn_buffers(args, client.comm, client.arena, online,
                              work=client.work)
 
@@ -215,7 +232,8 @@ def main():
             'data': 'synthetic',
             'config': {'model': b.model, 'global_batch': world * b.batch,
                        'seq_len': None,
-                       'parallelism': 'fedavg_dp%d_tau%d' % (world, TAU)},
+                       'parallelism': '%s_dp%d_tau%d' % (b.algo, world,
+                                                         TAU)},
         }
         print(json.dumps(out), flush=True)
     if dist.is_initialized():
