@@ -726,10 +726,13 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                  "bn_fwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                  __hip_bfloat16, scalar_t>;
-    hipLaunchKernelGGL(bn_stats_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+    const bool vec = (HW % BnVec<T>::N) == 0;
+    hipLaunchKernelGGL(vec ? bn_stats_kernel_v<T> : bn_stats_kernel<T>,
+                       grid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(x.data_ptr()), N, C, HW,
                        stats.data_ptr<float>());
-    hipLaunchKernelGGL(bn_norm_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+    hipLaunchKernelGGL(vec ? bn_norm_kernel_v<T> : bn_norm_kernel<T>,
+                       grid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(x.data_ptr()),
                        reinterpret_cast<T*>(y.data_ptr()),
                        stats.data_ptr<float>(),
@@ -761,13 +764,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                  __hip_bfloat16, scalar_t>;
     const T* yp = relu ? reinterpret_cast<const T*>(y.data_ptr()) : nullptr;
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<T>, grid, dim3(FT_BLOCK), 0,
+    const bool vec = (HW % BnVec<T>::N) == 0;
+    hipLaunchKernelGGL(vec ? bn_bwd_stats_kernel_v<T>
+                           : bn_bwd_stats_kernel<T>, grid, dim3(FT_BLOCK), 0,
                        STREAM, reinterpret_cast<const T*>(dy.data_ptr()),
                        reinterpret_cast<const T*>(x.data_ptr()), yp,
                        save_mean.data_ptr<float>(),
                        save_ivar.data_ptr<float>(), N, C, HW,
                        red.data_ptr<float>(), relu ? 1 : 0);
-    hipLaunchKernelGGL(bn_bwd_dx_kernel<T>, grid, dim3(FT_BLOCK), 0, STREAM,
+    hipLaunchKernelGGL(vec ? bn_bwd_dx_kernel_v<T> : bn_bwd_dx_kernel<T>,
+                       grid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(dy.data_ptr()),
                        reinterpret_cast<const T*>(x.data_ptr()), yp,
                        red.data_ptr<float>(), save_mean.data_ptr<float>(),
