@@ -40,7 +40,7 @@ def parse():
     p.add_argument('--gpus', type=int, default=1)
     p.add_argument('--steps', type=int, default=100)
     p.add_argument('--warmup', type=int, default=20)
-    p.add_argument('--batch', type=int, default=128)
+    p.add_argument('--batch', type=int, default=256)
     p.add_argument('--model', type=str, default='resnet20')
     p.add_argument('--dtype', type=str, default='bf16')
     p.add_argument('--graph', type=str, default='auto',
