@@ -7,7 +7,7 @@ import math
 
 import torch
 
-from fedtorch_amd import ops
+
 
 
 class FusedAdamW(object):

@@ -14,7 +14,7 @@ import torch
 
 from fedtorch_amd.components.comps import create_components
 from fedtorch_amd.components.optimizer import define_optimizer
-from fedtorch_amd.components.dataset import define_dataset, _load_data_batch
+from fedtorch_amd.components.dataset import define_dataset
 from fedtorch_amd.parallel.arena import Arena
 from fedtorch_amd.utils.init_config import init_config_centered
 from fedtorch_amd.logs.logging import configure_log, log_args

@@ -12,8 +12,7 @@ import torch
 
 from fedtorch_amd.components.scheduler import adjust_learning_rate
 from fedtorch_amd.components.dataset import load_data_batch
-from fedtorch_amd.trainings.flow import (
-    get_current_epoch, get_current_local_step, is_sync_fed)
+from fedtorch_amd.trainings.flow import get_current_epoch, is_sync_fed
 from fedtorch_amd.trainings.eval import inference
 from fedtorch_amd.trainings.federated import amp
 from fedtorch_amd.trainings.eval_centered import (

@@ -17,8 +17,7 @@ import torch
 
 from fedtorch_amd.components.scheduler import adjust_learning_rate
 from fedtorch_amd.components.dataset import load_data_batch
-from fedtorch_amd.trainings.flow import (
-    get_current_epoch, get_current_local_step, is_sync_fed)
+from fedtorch_amd.trainings.flow import get_current_epoch, is_sync_fed
 from fedtorch_amd.trainings.eval import inference, do_validate
 from fedtorch_amd.aggregation.federated import (
     fedavg_aggregation, fedgate_aggregation, scaffold_aggregation,
