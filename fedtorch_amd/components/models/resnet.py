@@ -12,6 +12,8 @@ padding games) so MIOpen picks NHWC kernels under bf16 autocast on gfx950.
 import torch
 import torch.nn as nn
 
+from fedtorch_amd.ops.batchnorm import BNReLU
+
 _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10, 'mnist': 10,
                 'fashion_mnist': 10, 'emnist': 10, 'emnist_full': 62,
                 'stl10': 10, 'imagenet': 1000}
@@ -35,7 +37,7 @@ class BasicBlock(nn.Module):
     def __init__(self, inplanes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = conv3x3(inplanes, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = BNReLU(planes)  # BN+ReLU fused on GPU (ops/batchnorm.py)
         self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
         self.bn2 = nn.BatchNorm2d(planes)
@@ -43,7 +45,7 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -56,9 +58,9 @@ class Bottleneck(nn.Module):
     def __init__(self, inplanes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(inplanes, planes, kernel_size=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = BNReLU(planes)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = BNReLU(planes)
         self.conv3 = nn.Conv2d(planes, planes * 4, kernel_size=1, bias=False)
         self.bn3 = nn.BatchNorm2d(planes * 4)
         self.relu = nn.ReLU(inplace=True)
@@ -66,8 +68,8 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -122,7 +124,7 @@ class ResNetCifar(_ResNetBase):
         self.num_classes = _num_classes(dataset)
         self.inplanes = 16
         self.conv1 = conv3x3(3, 16)
-        self.bn1 = nn.BatchNorm2d(16)
+        self.bn1 = BNReLU(16)
         self.relu = nn.ReLU(inplace=True)
         self.layer1 = self._make_stage(block_fn, 16, block_num)
         self.layer2 = self._make_stage(block_fn, 32, block_num, stride=2)
@@ -132,7 +134,7 @@ class ResNetCifar(_ResNetBase):
         self._init_weights()
 
     def forward(self, x):
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = self.bn1(self.conv1(x))
         x = self.layer3(self.layer2(self.layer1(x)))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)
@@ -150,7 +152,7 @@ class ResNetImageNet(_ResNetBase):
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2, padding=3,
                                bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = BNReLU(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
         self.layer1 = self._make_stage(block_fn, 64, block_nums[0])
@@ -162,7 +164,7 @@ class ResNetImageNet(_ResNetBase):
         self._init_weights()
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

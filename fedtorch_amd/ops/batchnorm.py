@@ -16,38 +16,50 @@ from fedtorch_amd import ops
 class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
-                eps):
+                eps, relu):
         empty = torch.empty(0, device=x.device)
         y, save_mean, save_ivar = ops._C.bn_fwd_train(
             x, weight if weight is not None else empty,
             bias if bias is not None else empty,
             running_mean if running_mean is not None else empty,
             running_var if running_var is not None else empty,
-            float(eps), float(momentum), False)
-        ctx.save_for_backward(x, weight, save_mean, save_ivar)
+            float(eps), float(momentum), bool(relu))
+        ctx.relu = bool(relu)
+        if relu:
+            ctx.save_for_backward(x, weight, save_mean, save_ivar, y)
+        else:
+            ctx.save_for_backward(x, weight, save_mean, save_ivar)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, save_mean, save_ivar = ctx.saved_tensors
+        if ctx.relu:
+            x, weight, save_mean, save_ivar, y = ctx.saved_tensors
+        else:
+            x, weight, save_mean, save_ivar = ctx.saved_tensors
+            y = torch.empty(0, device=x.device)
         empty = torch.empty(0, device=x.device)
         dx, dweight, dbias = ops._C.bn_bwd(
-            dy.contiguous(), x, empty, save_mean, save_ivar,
-            weight if weight is not None else empty, False)
+            dy.contiguous(), x, y, save_mean, save_ivar,
+            weight if weight is not None else empty, ctx.relu)
         return (dx, dweight if weight is not None else None,
                 dbias if weight is not None else None, None, None, None,
-                None)
+                None, None)
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
     """Drop-in BatchNorm2d: fused HIP kernels for GPU training, stock path
-    otherwise."""
+    otherwise.  ``fuse_relu=True`` folds the following ReLU into the same
+    kernels (fwd clamp + bwd mask on the saved output)."""
+
+    fuse_relu = False
 
     def forward(self, x):
         use_fused = (self.training and x.is_cuda and x.dim() == 4
                      and ops.hip_available() and not ops.FORCE_EAGER)
         if not use_fused:
-            return super().forward(x)
+            y = super().forward(x)
+            return torch.relu(y) if self.fuse_relu else y
         if self.track_running_stats and self.num_batches_tracked is not None:
             self.num_batches_tracked.add_(1)
         momentum = self.momentum if self.momentum is not None else 0.1
@@ -55,25 +67,49 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             x.contiguous(), self.weight, self.bias,
             self.running_mean if self.track_running_stats else None,
             self.running_var if self.track_running_stats else None,
-            momentum, self.eps)
+            momentum, self.eps, self.fuse_relu)
+
+
+class BNReLU(nn.Module):
+    """BatchNorm2d + ReLU as one module so the fused-BN conversion can fold
+    the ReLU into the BN kernels (`convert_to_fused_bn`)."""
+
+    def __init__(self, planes):
+        super().__init__()
+        self.bn = nn.BatchNorm2d(planes)
+
+    def forward(self, x):
+        y = self.bn(x)
+        # when the inner BN is the fused kernel with fuse_relu, the clamp
+        # already happened (checking the flag here keeps the module
+        # deepcopy-safe — no bound-method monkeypatching).
+        if getattr(self.bn, 'fuse_relu', False):
+            return y
+        return torch.relu(y)
 
 
 def convert_to_fused_bn(module):
     """Recursively replace nn.BatchNorm2d with FusedBatchNorm2d (params and
     buffers are re-used in place, state_dict layout unchanged)."""
+    def _fused_of(child, relu):
+        fused = FusedBatchNorm2d(
+            child.num_features, eps=child.eps, momentum=child.momentum,
+            affine=child.affine,
+            track_running_stats=child.track_running_stats)
+        fused.fuse_relu = relu
+        fused.weight = child.weight
+        fused.bias = child.bias
+        if child.track_running_stats:
+            fused.running_mean = child.running_mean
+            fused.running_var = child.running_var
+            fused.num_batches_tracked = child.num_batches_tracked
+        return fused
+
     for name, child in module.named_children():
         if type(child) is nn.BatchNorm2d:
-            fused = FusedBatchNorm2d(
-                child.num_features, eps=child.eps, momentum=child.momentum,
-                affine=child.affine,
-                track_running_stats=child.track_running_stats)
-            fused.weight = child.weight
-            fused.bias = child.bias
-            if child.track_running_stats:
-                fused.running_mean = child.running_mean
-                fused.running_var = child.running_var
-                fused.num_batches_tracked = child.num_batches_tracked
-            setattr(module, name, fused)
+            setattr(module, name, _fused_of(child, relu=False))
+        elif type(child) is BNReLU:
+            child.bn = _fused_of(child.bn, relu=True)
         else:
             convert_to_fused_bn(child)
     return module

@@ -165,3 +165,21 @@ def test_arena_packs_bn_running_stats():
     # no-BN model: no buffer arena
     arena2 = Arena(nn.Linear(4, 2))
     assert arena2.buf_flat is None
+
+
+def test_bnrelu_deepcopy_safe():
+    """APFL/PerFedMe deepcopy converted models; the copy's BN must be its
+    OWN module (no bound-method aliasing back to the original)."""
+    from copy import deepcopy
+    from fedtorch_amd.ops.batchnorm import BNReLU, convert_to_fused_bn
+    m = nn.Sequential(nn.Conv2d(3, 4, 3), BNReLU(4))
+    convert_to_fused_bn(m)
+    c = deepcopy(m)
+    assert c[1].bn is not m[1].bn
+    assert c[1].bn.weight is not m[1].bn.weight
+    x = torch.randn(2, 3, 8, 8)
+    m.eval()
+    c.eval()
+    assert torch.allclose(m(x), c(x))
+    # relu applied exactly once on the fallback path
+    assert (c(x) >= 0).all()

@@ -262,3 +262,28 @@ def test_fused_batchnorm_bf16_io():
     ref = nn.BatchNorm2d(16).cuda()
     yr = ref(x.float())
     assert torch.allclose(y.float(), yr, atol=0.05)
+
+
+def test_fused_bn_relu_matches_torch():
+    import torch.nn as nn
+    from fedtorch_amd.ops.batchnorm import FusedBatchNorm2d
+    torch.manual_seed(22)
+    N, C, H, W = 16, 32, 8, 8
+    x = torch.randn(N, C, H, W, device='cuda')
+    ref = nn.BatchNorm2d(C).cuda()
+    ref.weight.data.uniform_(0.5, 1.5)
+    ref.bias.data.uniform_(-0.5, 0.5)
+    fus = FusedBatchNorm2d(C).cuda()
+    fus.fuse_relu = True
+    fus.load_state_dict(ref.state_dict())
+    xr = x.clone().requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    yr = torch.relu(ref(xr))
+    yf = fus(xf)
+    assert torch.allclose(yf, yr, atol=2e-5)
+    g = torch.randn_like(yr)
+    yr.backward(g)
+    yf.backward(g)
+    assert torch.allclose(xf.grad, xr.grad, atol=2e-4)
+    assert torch.allclose(fus.weight.grad, ref.weight.grad, atol=1e-3)
+    assert torch.allclose(fus.bias.grad, ref.bias.grad, atol=1e-3)
