@@ -130,3 +130,84 @@ def test_fedavg_equals_mean_of_clients():
         p.join(timeout=120)
         assert p.exitcode == 0
     assert all(ok for _, ok in results)
+
+
+def _worker_algo(rank, world, port, algo, q):
+    try:
+        os.environ['MASTER_ADDR'] = '127.0.0.1'
+        os.environ['MASTER_PORT'] = str(port)
+        os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+        dist.init_process_group('gloo', rank=rank, world_size=world)
+        from fedtorch_amd.parameters import get_args
+        from fedtorch_amd.main import main as dispatch_main
+        argv = ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+                '--num_comms', '2', '--online_client_rate', '1.0',
+                '-b', '25', '--lr', '0.1', '--on_cuda', 'false',
+                '--dist_backend', 'gloo', '-j', '0',
+                '--checkpoint', '/tmp/ft_ci_%s' % algo, '--debug', 'false',
+                '--manual_seed', '11']
+        if algo == 'drfa':
+            argv += ['--federated_drfa', 'true', '--federated_type',
+                     'fedavg', '--local_step', '3',
+                     '--federated_sync_type', 'local_step']
+        else:
+            argv += ['--federated_type', algo]
+        if algo == 'apfl':
+            argv += ['--fed_adaptive_alpha', 'true']
+        args = get_args(argv)
+        dispatch_main(args)
+        import torch as _t
+        q.put((rank, True))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception:
+        import traceback
+        traceback.print_exc()
+        q.put((rank, False))
+        raise
+
+
+@pytest.mark.parametrize('algo', ['apfl', 'afl', 'drfa', 'packed'])
+def test_two_proc_algo_loops(algo):
+    world = 2
+    port = 29760 + abs(hash(algo)) % 150
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    target = _worker_packed if algo == 'packed' else _worker_algo
+    args_t = (world, port, algo, q) if algo != 'packed' else (world, port, q)
+    procs = [ctx.Process(target=target, args=(r,) + args_t)
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get() for _ in range(world)]
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    assert all(ok for _, ok in results)
+
+
+def _worker_packed(rank, world, port, q):
+    try:
+        os.environ['MASTER_ADDR'] = '127.0.0.1'
+        os.environ['MASTER_PORT'] = str(port)
+        os.environ['FEDTORCH_SYNTH_SIZE'] = '300'
+        dist.init_process_group('gloo', rank=rank, world_size=world)
+        from fedtorch_amd.parameters import get_args
+        from fedtorch_amd.main import main as dispatch_main
+        args = get_args([
+            '-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+            '--federated_type', 'fedavg', '--num_comms', '2',
+            '--online_client_rate', '1.0', '--local_step', '2',
+            '--federated_sync_type', 'local_step', '-b', '25',
+            '--lr', '0.1', '--on_cuda', 'false', '--dist_backend', 'gloo',
+            '-j', '0', '--clients_per_rank', '3', '--in_momentum', 'true',
+            '--checkpoint', '/tmp/ft_ci_packed', '--debug', 'false'])
+        dispatch_main(args)
+        q.put((rank, True))
+        dist.barrier()
+        dist.destroy_process_group()
+    except Exception:
+        import traceback
+        traceback.print_exc()
+        q.put((rank, False))
+        raise
