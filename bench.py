@@ -54,6 +54,13 @@ def parse():
                    choices=['on', 'off'])
     p.add_argument('--fp32_stem', type=str, default='off',
                    choices=['on', 'off'])
+    p.add_argument('--clients', type=int, default=1,
+                   help='virtual clients per rank (packed mode when > 1: '
+                        'replicas resident in HBM, BASELINE config 5)')
+    p.add_argument('--streams', type=int, default=4,
+                   help='concurrent HIP streams (packed mode): clients '
+                        'time-share this many model replicas, each with '
+                        'its own captured hipGraph')
     p.add_argument('--algo', type=str, default='fedavg',
                    choices=['fedavg', 'comgate_topk', 'comgate_quant'],
                    help='sync algorithm: plain FedAvg or FedCOMGATE '
@@ -128,6 +135,10 @@ def main():
         pg['lr'] = lr
     online = list(range(world))
     client.model.train()
+
+    if b.clients > 1:
+        return run_packed_bench(b, client, args, xs, ys, pool_n, world,
+                                rank, on_gpu)
 
     def inner(x, y):
         """one local SGD step: fwd + loss + bwd + fused arena step.
@@ -234,6 +245,157 @@ def main():
                        'seq_len': None,
                        'parallelism': '%s_dp%d_tau%d' % (b.algo, world,
                                                          TAU)},
+        }
+        print(json.dumps(out), flush=True)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+
+def run_packed_bench(b, client, args, xs, ys, pool_n, world, rank, on_gpu):
+    """Packed virtual clients: C client states per rank, K = --streams
+    model replicas executing CONCURRENTLY on separate HIP streams (each
+    slot replays its own captured graph).  Aggregation = one batched
+    multi_diff_accumulate over the [C, N] replica arena + one all-reduce.
+    """
+    from copy import deepcopy
+    from fedtorch_amd.parallel.arena import Arena
+    from fedtorch_amd.components.optim.sgd import FusedSGD
+    from fedtorch_amd import ops as ft_ops
+
+    C = b.clients
+    K = min(b.streams, C) if on_gpu else 1
+    device = client.arena.flat.device
+    n = client.arena.numel
+    g0 = client.optimizer.param_groups[0]
+
+    # per-client resident state
+    state = torch.zeros((C, n), device=device)
+    mom = torch.zeros((C, n), device=device)
+    for c in range(C):
+        state[c].copy_(client.arena.flat)
+    server = client.arena.clone_flat()
+    weights = torch.full((C,), 1.0 / (C * world), device=device)
+    partial = torch.zeros(n, device=device)
+
+    # K execution slots: independent module replicas + arenas + optimizers
+    slots = []
+    for k in range(K):
+        m = deepcopy(client.model)
+        arena = Arena(m)
+        opt = FusedSGD(arena, lr=g0['lr'], in_momentum=g0['in_momentum'],
+                       weight_decay=g0['weight_decay'])
+        sx = (xs[0] if isinstance(xs, list) else xs[0]).clone()
+        sy = ys[0].clone()
+        stream = torch.cuda.Stream() if on_gpu else None
+        slots.append(dict(model=m, arena=arena, opt=opt, sx=sx, sy=sy,
+                          stream=stream, graph=None))
+
+    def slot_step(sl):
+        sl['opt'].zero_grad()
+        with amp(args):
+            loss = client.criterion(sl['model'](sl['sx']), sl['sy'])
+        loss.backward()
+        sl['opt'].step(apply_lr=True, apply_in_momentum=True,
+                       apply_out_momentum=False)
+
+    if on_gpu and b.graph != 'off':
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for sl in slots:
+                for _ in range(3):
+                    slot_step(sl)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        for sl in slots:
+            gph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gph):
+                slot_step(sl)
+            sl['graph'] = gph
+
+    main_stream = torch.cuda.current_stream() if on_gpu else None
+
+    # hipGraph + momentum caveat: a captured graph bakes in the slot's
+    # momentum buffer POINTER, so per-client momentum rows are STAGED into
+    # the slot buffer around each client's tau steps (copies are ~N floats,
+    # trivial next to the steps themselves).
+    for sl in slots:
+        if sl['opt']._in_buf is None:
+            sl['opt'].bind_state(in_buf=torch.zeros(n, device=device),
+                                 in_init=True)
+
+    def run_round_staged(r):
+        for base in range(0, C, K):
+            group = list(range(base, min(base + K, C)))
+            if on_gpu:
+                for k, c in enumerate(group):
+                    sl = slots[k]
+                    sl['stream'].wait_stream(main_stream)
+                    with torch.cuda.stream(sl['stream']):
+                        sl['arena'].flat.copy_(server)
+                        sl['opt']._in_buf.copy_(mom[c])
+                        for t in range(TAU):
+                            sl['sx'].copy_(xs[(c + t) % pool_n])
+                            sl['sy'].copy_(ys[(c + t) % pool_n])
+                            if sl['graph'] is not None:
+                                sl['graph'].replay()
+                            else:
+                                slot_step(sl)
+                        state[c].copy_(sl['arena'].flat)
+                        mom[c].copy_(sl['opt']._in_buf)
+                    main_stream.wait_stream(sl['stream'])
+            else:
+                for k, c in enumerate(group):
+                    sl = slots[k]
+                    sl['arena'].flat.copy_(server)
+                    sl['opt']._in_buf.copy_(mom[c])
+                    for t in range(TAU):
+                        sl['sx'].copy_(xs[(c + t) % pool_n])
+                        sl['sy'].copy_(ys[(c + t) % pool_n])
+                        slot_step(sl)
+                    state[c].copy_(sl['arena'].flat)
+                    mom[c].copy_(sl['opt']._in_buf)
+        ft_ops.multi_diff_accumulate(server, state, weights, partial)
+        client.comm.all_reduce(partial)
+        server.sub_(partial)
+
+    rounds_warm = max(b.warmup // TAU, 1)
+    rounds_timed = max(b.steps // TAU, 1)
+    for r in range(rounds_warm):
+        run_round_staged(r)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for r in range(rounds_timed):
+        run_round_staged(r + rounds_warm)
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+    if rank == 0:
+        total_steps = rounds_timed * TAU * C
+        value = world * b.batch * total_steps / elapsed
+        out = {
+            'metric': 'samples/sec', 'value': value, 'unit': 'samples/sec',
+            'n_gpus': world, 'steps': rounds_timed * TAU,
+            'warmup': rounds_warm * TAU,
+            'ms_per_step': elapsed / (rounds_timed * TAU) * 1e3,
+            'higher_is_better': True, 'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'bf16' if args.bf16 else 'fp32', 'data': 'synthetic',
+            'config': {'model': b.model,
+                       'global_batch': world * b.batch * C,
+                       'seq_len': None,
+                       'parallelism': 'fedavg_packed_c%d_k%d_dp%d_tau%d' % (
+                           C, K, world, TAU)},
         }
         print(json.dumps(out), flush=True)
     if dist.is_initialized():
