@@ -149,6 +149,16 @@ def train_and_validate_federated_centered(Clients, Server):
                 Clients[oc].args, Server.model, Server.criterion,
                 Server.metrics, Server.optimizer, Clients[oc].train_loader,
                 Server.global_val_tracker, val=False, local=False)
+            if args.per_class_acc:
+                # per-client trackers feed the worst/best/var lines
+                # (reference `centered/main.py:77-88`)
+                Clients[oc].reset_tracker(Clients[oc].local_val_tracker)
+                Clients[oc].reset_tracker(Clients[oc].global_val_tracker)
+                do_validate_centered(
+                    Clients[oc].args, Server.model, Server.criterion,
+                    Server.metrics, Server.optimizer,
+                    Clients[oc].train_loader,
+                    Clients[oc].global_val_tracker, val=False, local=False)
             if args.fed_personal:
                 do_validate_centered(
                     Clients[oc].args, Server.model, Server.criterion,
@@ -174,6 +184,12 @@ def train_and_validate_federated_centered(Clients, Server):
                 Clients[oc].metrics, Clients[oc].optimizer,
                 Clients[oc].train_loader, Server.local_val_tracker,
                 val=False, local=True)
+            if args.per_class_acc:
+                do_validate_centered(
+                    Clients[oc].args, Clients[oc].model,
+                    Clients[oc].criterion, Clients[oc].metrics,
+                    Clients[oc].optimizer, Clients[oc].train_loader,
+                    Clients[oc].local_val_tracker, val=False, local=True)
             if args.fed_personal:
                 do_validate_centered(
                     Clients[oc].args, Clients[oc].model,
