@@ -22,3 +22,12 @@ Public API parity with the reference (README.md:44-78 of the reference):
 __version__ = "0.1.0"
 
 from fedtorch_amd.parameters import get_args  # noqa: F401
+
+
+def __getattr__(name):
+    # lazy: `from fedtorch_amd import Client` without importing torch at
+    # package-import time
+    if name in ('Client', 'ClientCentered', 'ServerCentered'):
+        import fedtorch_amd.nodes as nodes
+        return getattr(nodes, name)
+    raise AttributeError(name)
