@@ -57,10 +57,11 @@ def parse():
     p.add_argument('--clients', type=int, default=1,
                    help='virtual clients per rank (packed mode when > 1: '
                         'replicas resident in HBM, BASELINE config 5)')
-    p.add_argument('--streams', type=int, default=4,
-                   help='concurrent HIP streams (packed mode): clients '
-                        'time-share this many model replicas, each with '
-                        'its own captured hipGraph')
+    p.add_argument('--streams', type=int, default=1,
+                   help='concurrent HIP streams (packed mode). Measured on '
+                        'MI355X: multi-stream replay does NOT overlap for '
+                        'this kernel mix (profiles/r01_bench_notes.md), so '
+                        'the default is 1.')
     p.add_argument('--algo', type=str, default='fedavg',
                    choices=['fedavg', 'comgate_topk', 'comgate_quant'],
                    help='sync algorithm: plain FedAvg or FedCOMGATE '
