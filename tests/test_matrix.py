@@ -19,6 +19,7 @@ CASES = {
     'robust_logreg': ['-d', 'mnist', '-a', 'robust_logistic_regression',
                       '-b', '32'],
     'cifar_densenet': ['-d', 'cifar10', '-a', 'densenet', '-b', '8'],
+    'cifar_resnet56': ['-d', 'cifar10', '-a', 'resnet56', '-b', '8'],
     'cifar_wideresnet': ['-d', 'cifar10', '-a', 'wideresnet', '-b', '8'],
     'mnist_dirichlet': ['-d', 'mnist', '-a', 'mlp', '--iid_data', 'false',
                         '--dirichlet', 'true', '-b', '16'],
