@@ -287,3 +287,26 @@ def test_fused_bn_relu_matches_torch():
     assert torch.allclose(xf.grad, xr.grad, atol=2e-4)
     assert torch.allclose(fus.weight.grad, ref.weight.grad, atol=1e-3)
     assert torch.allclose(fus.bias.grad, ref.bias.grad, atol=1e-3)
+
+
+def test_bench_contract():
+    """bench.py emits the driver-contract JSON line and runs the native
+    path (subprocess, tiny step count)."""
+    import json
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, 'bench.py'), '--steps', '12',
+         '--warmup', '4', '--batch', '64'],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith('{')][-1]
+    d = json.loads(line)
+    assert d['metric'] == 'samples/sec'
+    assert d['value'] > 0
+    assert d['n_gpus'] == 1
+    assert d['dtype'] == 'bf16'
+    assert d['data'] == 'synthetic'
+    assert d['config']['model'] == 'resnet20'
