@@ -12,7 +12,7 @@ padding games) so MIOpen picks NHWC kernels under bf16 autocast on gfx950.
 import torch
 import torch.nn as nn
 
-from fedtorch_amd.ops.batchnorm import BNReLU
+from fedtorch_amd.ops.batchnorm import BNReLU, BNAddReLU
 
 _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10, 'mnist': 10,
                 'fashion_mnist': 10, 'emnist': 10, 'emnist_full': 62,
@@ -38,18 +38,16 @@ class BasicBlock(nn.Module):
         super().__init__()
         self.conv1 = conv3x3(inplanes, planes, stride)
         self.bn1 = BNReLU(planes)  # BN+ReLU fused on GPU (ops/batchnorm.py)
-        self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        # block tail relu(bn2(conv2) + identity): add+ReLU fold into the BN
+        # kernels on GPU (ops/batchnorm.py BNAddReLU)
+        self.bn2 = BNAddReLU(planes)
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x
+        identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        return self.relu(out + identity)
+        return self.bn2(self.conv2(out), identity)
 
 
 class Bottleneck(nn.Module):
@@ -62,18 +60,14 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(planes, planes, stride)
         self.bn2 = BNReLU(planes)
         self.conv3 = nn.Conv2d(planes, planes * 4, kernel_size=1, bias=False)
-        self.bn3 = nn.BatchNorm2d(planes * 4)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn3 = BNAddReLU(planes * 4)
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x
+        identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        return self.relu(out + identity)
+        return self.bn3(self.conv3(out), identity)
 
 
 class _ResNetBase(nn.Module):
@@ -125,7 +119,6 @@ class ResNetCifar(_ResNetBase):
         self.inplanes = 16
         self.conv1 = conv3x3(3, 16)
         self.bn1 = BNReLU(16)
-        self.relu = nn.ReLU(inplace=True)
         self.layer1 = self._make_stage(block_fn, 16, block_num)
         self.layer2 = self._make_stage(block_fn, 32, block_num, stride=2)
         self.layer3 = self._make_stage(block_fn, 64, block_num, stride=2)
@@ -153,7 +146,6 @@ class ResNetImageNet(_ResNetBase):
         self.conv1 = nn.Conv2d(3, 64, kernel_size=7, stride=2, padding=3,
                                bias=False)
         self.bn1 = BNReLU(64)
-        self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
         self.layer1 = self._make_stage(block_fn, 64, block_nums[0])
         self.layer2 = self._make_stage(block_fn, 128, block_nums[1], stride=2)

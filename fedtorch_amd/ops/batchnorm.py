@@ -14,17 +14,23 @@ from fedtorch_amd import ops
 
 
 class _FusedBNFunction(torch.autograd.Function):
+    """y = [relu](bn(x) [+ res]); res fuses a ResNet residual add into the
+    normalize kernel (and its gradient — the ReLU-masked dy — into the dx
+    kernel)."""
+
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
-                eps, relu):
+                eps, relu, res):
         empty = torch.empty(0, device=x.device)
         y, save_mean, save_ivar = ops._C.bn_fwd_train(
             x, weight if weight is not None else empty,
             bias if bias is not None else empty,
             running_mean if running_mean is not None else empty,
             running_var if running_var is not None else empty,
-            float(eps), float(momentum), bool(relu))
+            float(eps), float(momentum), bool(relu),
+            res if res is not None else empty)
         ctx.relu = bool(relu)
+        ctx.has_res = res is not None
         if relu:
             ctx.save_for_backward(x, weight, save_mean, save_ivar, y)
         else:
@@ -39,12 +45,12 @@ class _FusedBNFunction(torch.autograd.Function):
             x, weight, save_mean, save_ivar = ctx.saved_tensors
             y = torch.empty(0, device=x.device)
         empty = torch.empty(0, device=x.device)
-        dx, dweight, dbias = ops._C.bn_bwd(
+        dx, dweight, dbias, dres = ops._C.bn_bwd(
             dy.contiguous(), x, y, save_mean, save_ivar,
-            weight if weight is not None else empty, ctx.relu)
+            weight if weight is not None else empty, ctx.relu, ctx.has_res)
         return (dx, dweight if weight is not None else None,
                 dbias if weight is not None else None, None, None, None,
-                None, None)
+                None, None, dres if ctx.has_res else None)
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
@@ -53,21 +59,40 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
     kernels (fwd clamp + bwd mask on the saved output)."""
 
     fuse_relu = False
+    _nbt_pending = 0  # lazy num_batches_tracked increments (flushed on save)
 
-    def forward(self, x):
+    def forward(self, x, res=None):
         use_fused = (self.training and x.is_cuda and x.dim() == 4
                      and ops.hip_available() and not ops.FORCE_EAGER)
         if not use_fused:
+            self._flush_nbt()
             y = super().forward(x)
+            if res is not None:
+                return torch.relu(y + res)
             return torch.relu(y) if self.fuse_relu else y
         if self.track_running_stats and self.num_batches_tracked is not None:
-            self.num_batches_tracked.add_(1)
+            # a GPU .add_(1) here is one extra kernel per BN layer per step
+            # (it shows in rocprof as CUDAFunctorOnSelf_add<long>); count on
+            # the host instead and flush into the buffer only when the
+            # state_dict is read (checkpoint / sync).
+            self._nbt_pending += 1
         momentum = self.momentum if self.momentum is not None else 0.1
         return _FusedBNFunction.apply(
             x.contiguous(), self.weight, self.bias,
             self.running_mean if self.track_running_stats else None,
             self.running_var if self.track_running_stats else None,
-            momentum, self.eps, self.fuse_relu)
+            momentum, self.eps,
+            self.fuse_relu or res is not None,
+            res.contiguous() if res is not None else None)
+
+    def _flush_nbt(self):
+        if self._nbt_pending and self.num_batches_tracked is not None:
+            self.num_batches_tracked.add_(self._nbt_pending)
+            self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
 
 
 class BNReLU(nn.Module):
@@ -86,6 +111,21 @@ class BNReLU(nn.Module):
         if getattr(self.bn, 'fuse_relu', False):
             return y
         return torch.relu(y)
+
+
+class BNAddReLU(nn.Module):
+    """``relu(bn(x) + res)`` — the ResNet block tail — as one module so the
+    fused-BN conversion can fold the residual add AND the ReLU into the BN
+    normalize/dx kernels (3 fewer kernels per block per step)."""
+
+    def __init__(self, planes):
+        super().__init__()
+        self.bn = nn.BatchNorm2d(planes)
+
+    def forward(self, x, res):
+        if isinstance(self.bn, FusedBatchNorm2d):
+            return self.bn(x, res=res)
+        return torch.relu(self.bn(x) + res)
 
 
 def convert_to_fused_bn(module):
@@ -110,6 +150,9 @@ def convert_to_fused_bn(module):
             setattr(module, name, _fused_of(child, relu=False))
         elif type(child) is BNReLU:
             child.bn = _fused_of(child.bn, relu=True)
+        elif type(child) is BNAddReLU:
+            # relu comes from the res= path at call time, not the flag
+            child.bn = _fused_of(child.bn, relu=False)
         else:
             convert_to_fused_bn(child)
     return module

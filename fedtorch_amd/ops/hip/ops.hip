@@ -699,10 +699,24 @@ void scatter_accumulate(torch::Tensor out, torch::Tensor vs,
 // ==========================================================================
 // fused spatial BatchNorm (training fwd/bwd) — kernels in batchnorm.h
 // ==========================================================================
-static inline int bn_chunks(long per_ch, long C) {
-  long by_work = (per_ch + FT_BLOCK * 8 - 1) / (FT_BLOCK * 8);
-  long cap = FT_MAX_BLOCKS / (C > 0 ? C : 1);
-  long c = by_work < cap ? by_work : cap;
+// number of per-channel partial blocks for the reduction kernels: target
+// ~512 blocks total (2/CU for a ~2-5 us kernel), ≤64 partials per channel
+// so the consumer's finalize loop stays trivial.
+static inline int bn_nparts(long per_v, long C) {
+  long target = 512 / (C > 0 ? C : 1);
+  long by_work = (per_v + FT_BLOCK - 1) / FT_BLOCK;
+  long B = target < by_work ? target : by_work;
+  if (B > 64) B = 64;
+  if (B < 1) B = 1;
+  return (int)B;
+}
+
+// grid.x for the elementwise kernels (pure streaming, no partial buffer):
+// ~1024 blocks total, 2+ vector-iterations per thread.
+static inline int bn_ew_chunks(long per_v, long C) {
+  long cap = 1024 / (C > 0 ? C : 1);
+  long by_work = (per_v + FT_BLOCK - 1) / FT_BLOCK;
+  long c = cap < by_work ? cap : by_work;
   return (int)(c < 1 ? 1 : c);
 }
 
@@ -711,37 +725,45 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor running_mean,
                                         torch::Tensor running_var,
                                         double eps, double momentum,
-                                        bool relu) {
+                                        bool relu, torch::Tensor res) {
   CHK(x);
   TORCH_CHECK(x.dim() == 4, "bn_fwd_train expects NCHW");
   long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(N * HW < (1L << 31), "bn_fwd_train: N*HW must fit in int32");
   auto f32 = x.options().dtype(torch::kFloat);
-  auto stats = torch::zeros({C, 2}, f32);
   auto save_mean = torch::empty({C}, f32);
   auto save_ivar = torch::empty({C}, f32);
   auto y = torch::empty_like(x);
-  dim3 grid(bn_chunks(N * HW, C), C);
   bool track = running_mean.numel() > 0;
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
                                  "bn_fwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                  __hip_bfloat16, scalar_t>;
     const bool vec = (HW % BnVec<T>::N) == 0;
-    hipLaunchKernelGGL(vec ? bn_stats_kernel_v<T> : bn_stats_kernel<T>,
-                       grid, dim3(FT_BLOCK), 0, STREAM,
+    const int VN = vec ? BnVec<T>::N : 1;
+    const long per_v = (N * HW) / VN;
+    const int B = bn_nparts(per_v, C);
+    auto part = torch::empty({C, B, 2}, f32);
+    dim3 rgrid(B, C), egrid(bn_ew_chunks(per_v, C), C);
+    auto stats_fn = vec ? bn_stats_k<T, typename BnVec<T>::V, BnVec<T>::N>
+                        : bn_stats_k<T, typename Bn1<T>::V, 1>;
+    auto norm_fn = vec ? bn_norm_k<T, typename BnVec<T>::V, BnVec<T>::N>
+                       : bn_norm_k<T, typename Bn1<T>::V, 1>;
+    hipLaunchKernelGGL(stats_fn, rgrid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(x.data_ptr()), N, C, HW,
-                       stats.data_ptr<float>());
-    hipLaunchKernelGGL(vec ? bn_norm_kernel_v<T> : bn_norm_kernel<T>,
-                       grid, dim3(FT_BLOCK), 0, STREAM,
+                       part.data_ptr<float>());
+    hipLaunchKernelGGL(norm_fn, egrid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(x.data_ptr()),
                        reinterpret_cast<T*>(y.data_ptr()),
-                       stats.data_ptr<float>(),
+                       part.data_ptr<float>(), B,
                        weight.numel() ? weight.data_ptr<float>() : nullptr,
                        bias.numel() ? bias.data_ptr<float>() : nullptr,
                        save_mean.data_ptr<float>(),
                        save_ivar.data_ptr<float>(),
                        track ? running_mean.data_ptr<float>() : nullptr,
                        track ? running_var.data_ptr<float>() : nullptr,
+                       res.numel() ? reinterpret_cast<const T*>(res.data_ptr())
+                                   : nullptr,
                        N, C, HW, (float)eps, (float)momentum, relu ? 1 : 0);
   });
   return {y, save_mean, save_ivar};
@@ -750,40 +772,51 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor y, torch::Tensor save_mean,
                                   torch::Tensor save_ivar,
-                                  torch::Tensor weight, bool relu) {
+                                  torch::Tensor weight, bool relu,
+                                  bool has_res) {
   CHK(dy); CHK(x);
   long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(N * HW < (1L << 31), "bn_bwd: N*HW must fit in int32");
   auto f32 = x.options().dtype(torch::kFloat);
-  auto red = torch::zeros({C, 2}, f32);
   auto dx = torch::empty_like(x);
+  auto dres = has_res ? torch::empty_like(x)
+                      : torch::empty({0}, x.options());
   auto dweight = torch::empty({C}, f32);
   auto dbias = torch::empty({C}, f32);
-  dim3 grid(bn_chunks(N * HW, C), C);
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
                                  "bn_bwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                  __hip_bfloat16, scalar_t>;
     const T* yp = relu ? reinterpret_cast<const T*>(y.data_ptr()) : nullptr;
     const bool vec = (HW % BnVec<T>::N) == 0;
-    hipLaunchKernelGGL(vec ? bn_bwd_stats_kernel_v<T>
-                           : bn_bwd_stats_kernel<T>, grid, dim3(FT_BLOCK), 0,
+    const int VN = vec ? BnVec<T>::N : 1;
+    const long per_v = (N * HW) / VN;
+    const int B = bn_nparts(per_v, C);
+    auto part = torch::empty({C, B, 2}, f32);
+    dim3 rgrid(B, C), egrid(bn_ew_chunks(per_v, C), C);
+    auto stats_fn = vec ? bn_bwd_stats_k<T, typename BnVec<T>::V, BnVec<T>::N>
+                        : bn_bwd_stats_k<T, typename Bn1<T>::V, 1>;
+    auto dx_fn = vec ? bn_bwd_dx_k<T, typename BnVec<T>::V, BnVec<T>::N>
+                     : bn_bwd_dx_k<T, typename Bn1<T>::V, 1>;
+    hipLaunchKernelGGL(stats_fn, rgrid, dim3(FT_BLOCK), 0,
                        STREAM, reinterpret_cast<const T*>(dy.data_ptr()),
                        reinterpret_cast<const T*>(x.data_ptr()), yp,
                        save_mean.data_ptr<float>(),
                        save_ivar.data_ptr<float>(), N, C, HW,
-                       red.data_ptr<float>(), relu ? 1 : 0);
-    hipLaunchKernelGGL(vec ? bn_bwd_dx_kernel_v<T> : bn_bwd_dx_kernel<T>,
-                       grid, dim3(FT_BLOCK), 0, STREAM,
+                       part.data_ptr<float>(), relu ? 1 : 0);
+    hipLaunchKernelGGL(dx_fn, egrid, dim3(FT_BLOCK), 0, STREAM,
                        reinterpret_cast<const T*>(dy.data_ptr()),
                        reinterpret_cast<const T*>(x.data_ptr()), yp,
-                       red.data_ptr<float>(), save_mean.data_ptr<float>(),
+                       part.data_ptr<float>(), B, save_mean.data_ptr<float>(),
                        save_ivar.data_ptr<float>(),
                        weight.numel() ? weight.data_ptr<float>() : nullptr,
                        reinterpret_cast<T*>(dx.data_ptr()),
+                       has_res ? reinterpret_cast<T*>(dres.data_ptr())
+                               : nullptr,
                        dweight.data_ptr<float>(), dbias.data_ptr<float>(),
                        N, C, HW, relu ? 1 : 0);
   });
-  return {dx, dweight, dbias};
+  return {dx, dweight, dbias, dres};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

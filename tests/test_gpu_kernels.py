@@ -289,6 +289,51 @@ def test_fused_bn_relu_matches_torch():
     assert torch.allclose(fus.bias.grad, ref.bias.grad, atol=1e-3)
 
 
+@pytest.mark.parametrize('hw', [8, 7])  # 8x8 → vectorized, 7x7 → scalar path
+def test_fused_bn_add_relu_matches_torch(hw):
+    """BNAddReLU: relu(bn(x) + res) fused fwd + dres bwd vs eager."""
+    import torch.nn as nn
+    from fedtorch_amd.ops.batchnorm import BNAddReLU, convert_to_fused_bn
+    torch.manual_seed(23)
+    N, C = 16, 32
+    x = torch.randn(N, C, hw, hw, device='cuda')
+    res = torch.randn(N, C, hw, hw, device='cuda')
+    ref = nn.BatchNorm2d(C).cuda()
+    ref.weight.data.uniform_(0.5, 1.5)
+    ref.bias.data.uniform_(-0.5, 0.5)
+    fus = BNAddReLU(C).cuda()
+    fus.bn.load_state_dict(ref.state_dict())
+    convert_to_fused_bn(fus)
+    xr = x.clone().requires_grad_(True)
+    rr = res.clone().requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    rf = res.clone().requires_grad_(True)
+    yr = torch.relu(ref(xr) + rr)
+    yf = fus(xf, rf)
+    assert torch.allclose(yf, yr, atol=2e-5)
+    g = torch.randn_like(yr)
+    yr.backward(g)
+    yf.backward(g)
+    assert torch.allclose(xf.grad, xr.grad, atol=2e-4)
+    assert torch.allclose(rf.grad, rr.grad, atol=2e-5)
+    assert torch.allclose(fus.bn.weight.grad, ref.weight.grad, atol=1e-3)
+    assert torch.allclose(fus.bn.bias.grad, ref.bias.grad, atol=1e-3)
+    assert torch.allclose(fus.bn.running_mean, ref.running_mean, atol=1e-5)
+
+
+def test_fused_bn_nbt_lazy_flush():
+    """num_batches_tracked counts on the host and flushes into the buffer
+    when the state_dict is read (no per-step GPU kernel)."""
+    from fedtorch_amd.ops.batchnorm import FusedBatchNorm2d
+    bn = FusedBatchNorm2d(8).cuda()
+    x = torch.randn(4, 8, 8, 8, device='cuda')
+    for _ in range(3):
+        bn(x)
+    sd = bn.state_dict()
+    assert int(sd['num_batches_tracked']) == 3
+    assert bn._nbt_pending == 0
+
+
 def test_bench_contract():
     """bench.py emits the driver-contract JSON line and runs the native
     path (subprocess, tiny step count)."""
