@@ -697,6 +697,48 @@ void scatter_accumulate(torch::Tensor out, torch::Tensor vs,
 }
 
 // ==========================================================================
+// grad gather: scattered autograd-owned grad buffers -> flat grad arena.
+//
+// With p.grad = None at hipGraph-capture time, AccumulateGrad STEALS the
+// produced tensor instead of launching one fp32 add per parameter per step
+// (~65 CUDAFunctor_add kernels/step on ResNet-20, ~300 us at b256 —
+// profiles/r01_bench_notes.md).  This single kernel then copies every
+// stolen buffer into the contiguous grad arena that the fused SGD step and
+// the grad-norm trackers read.  Chunk table is built once per capture on
+// the host (shapes are static).
+// ==========================================================================
+__global__ void gather_chunks_kernel(const unsigned long long* __restrict__
+                                         srcs,
+                                     const int* __restrict__ table /*[B,4]*/,
+                                     float* __restrict__ dst) {
+  const int* e = table + 4 * blockIdx.x;
+  const float* src =
+      reinterpret_cast<const float*>(srcs[e[0]]) + e[1];
+  float* out = dst + e[2];
+  const int n = e[3];
+  const int n4 = n >> 2;
+  for (int i = threadIdx.x; i < n4; i += blockDim.x)
+    reinterpret_cast<float4*>(out)[i] =
+        reinterpret_cast<const float4*>(src)[i];
+  for (int i = (n4 << 2) + threadIdx.x; i < n; i += blockDim.x)
+    out[i] = src[i];
+}
+
+void gather_grads(torch::Tensor srcs, torch::Tensor table,
+                  torch::Tensor dst) {
+  CHK(dst); CHK(srcs); CHK(table);
+  TORCH_CHECK(table.dim() == 2 && table.size(1) == 4, "table must be [B,4]");
+  TORCH_CHECK(srcs.scalar_type() == torch::kLong, "srcs must be int64 ptrs");
+  TORCH_CHECK(table.scalar_type() == torch::kInt, "table must be int32");
+  const int B = (int)table.size(0);
+  hipLaunchKernelGGL(gather_chunks_kernel, dim3(B), dim3(FT_BLOCK), 0,
+                     STREAM,
+                     reinterpret_cast<const unsigned long long*>(
+                         srcs.data_ptr<long>()),
+                     table.data_ptr<int>(), dst.data_ptr<float>());
+}
+
+// ==========================================================================
 // fused spatial BatchNorm (training fwd/bwd) — kernels in batchnorm.h
 // ==========================================================================
 // number of per-channel partial blocks for the reduction kernels: target
@@ -837,4 +879,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_compress", &topk_compress);
   m.def("scatter_accumulate", &scatter_accumulate);
   m.def("multi_diff_accumulate", &multi_diff_accumulate);
+  m.def("gather_grads", &gather_grads);
 }

@@ -107,9 +107,53 @@ class Arena(object):
         """Point every p.grad at its view of the grad arena."""
         for p, off, n in zip(self.params, self.offsets, self.numels):
             p.grad = self.grad[off:off + n].view(p.shape)
+        self._gather_state = None
 
     def zero_grad(self):
         self.grad.zero_()
+
+    # ---- stolen-grad gather (hipGraph path) --------------------------------
+    # With p.grad = None at capture time, AccumulateGrad STEALS the produced
+    # tensor (no per-parameter fp32 add kernel); gather_grads() then runs ONE
+    # HIP kernel copying every stolen buffer into the contiguous grad arena
+    # the fused SGD / trackers read.  detach_grads() before capture,
+    # gather_grads() after loss.backward() inside the capture.
+    _GATHER_CHUNK = 4096  # fp32 elements per block
+
+    def detach_grads(self):
+        for p in self.params:
+            p.grad = None
+        self._gather_state = None
+
+    def _build_gather_state(self):
+        srcs, table = [], []
+        for i, (p, off, n) in enumerate(zip(self.params, self.offsets,
+                                            self.numels)):
+            g = p.grad
+            if (g is None or not g.is_contiguous()
+                    or g.dtype != torch.float32):
+                return None  # fall back to per-tensor copies
+            srcs.append(g.data_ptr())
+            for c in range(0, n, self._GATHER_CHUNK):
+                table.append((i, c, off + c, min(self._GATHER_CHUNK, n - c)))
+        dev = self.flat.device
+        return (torch.tensor(srcs, dtype=torch.int64, device=dev),
+                torch.tensor(table, dtype=torch.int32, device=dev))
+
+    def gather_grads(self):
+        """One-kernel copy of the stolen per-parameter grads into the grad
+        arena.  Builds the chunk table on first use (static shapes)."""
+        if getattr(self, '_gather_state', None) is None:
+            self._gather_state = self._build_gather_state()
+            if self._gather_state is None:
+                self._gather_state = ()
+        if self._gather_state:
+            from fedtorch_amd import ops
+            ops._C.gather_grads(self._gather_state[0],
+                                self._gather_state[1], self.grad)
+        else:  # non-contiguous / non-fp32 stolen grads: plain copies
+            for p, off, n in zip(self.params, self.offsets, self.numels):
+                self.grad[off:off + n].copy_(p.grad.reshape(-1))
 
     # ---- flat state helpers ------------------------------------------------
     def new_buffer(self, zero=True):

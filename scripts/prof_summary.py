@@ -22,7 +22,7 @@ def summarize(db_path, out_path, top=30, window=0.0):
         "SELECT MIN(start), MAX(end) FROM rocpd_kernel_dispatch_%s"
         % sfx).fetchone()
     cut = t0 + (t1 - t0) * window
-    where = "WHERE k.start >= %d" % cut
+    where = "WHERE CAST(k.start AS REAL) >= %d" % cut
     tot = cur.execute(
         "SELECT SUM(k.end-k.start)/1e6, COUNT(*), "
         "(MAX(k.end)-MIN(k.start))/1e6 "

@@ -321,6 +321,42 @@ def test_fused_bn_add_relu_matches_torch(hw):
     assert torch.allclose(fus.bn.running_mean, ref.running_mean, atol=1e-5)
 
 
+def test_gather_grads_matches_attached():
+    """Stolen-grad gather (graph path) fills the grad arena with exactly
+    what attached-view accumulation produces."""
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    torch.manual_seed(24)
+
+    def make():
+        torch.manual_seed(7)
+        return nn.Sequential(nn.Linear(33, 65), nn.ReLU(),
+                             nn.Linear(65, 10)).cuda()
+
+    x = torch.randn(16, 33, device='cuda')
+    y = torch.randint(0, 10, (16,), device='cuda')
+    crit = nn.CrossEntropyLoss()
+
+    m1 = make()
+    a1 = Arena(m1)
+    a1.zero_grad()
+    crit(m1(x), y).backward()
+    ref = a1.grad.clone()
+
+    m2 = make()
+    a2 = Arena(m2)
+    a2.detach_grads()
+    crit(m2(x), y).backward()
+    a2.gather_grads()
+    torch.cuda.synchronize()
+    assert torch.allclose(a2.grad, ref, atol=1e-6)
+    # second call reuses the cached chunk table
+    assert a2._gather_state
+    a2.gather_grads()
+    torch.cuda.synchronize()
+    assert torch.allclose(a2.grad, ref, atol=1e-6)
+
+
 def test_fused_bn_nbt_lazy_flush():
     """num_batches_tracked counts on the host and flushes into the buffer
     when the state_dict is read (no per-step GPU kernel)."""
