@@ -23,15 +23,20 @@ def create_components(args):
     model = define_model(args)
     if args.graph.on_cuda and torch.cuda.is_available():
         model = model.cuda()
-        if getattr(args, 'channels_last', False):
+        cl = getattr(args, 'channels_last', False)
+        if cl:
             model = model.to(memory_format=torch.channels_last)
         if getattr(args, 'fused_bn', True) and \
-                getattr(args, 'hip_kernels', True) and \
-                not getattr(args, 'channels_last', False):
+                getattr(args, 'hip_kernels', True):
             from fedtorch_amd.ops.batchnorm import convert_to_fused_bn
             import fedtorch_amd.ops as _ops
             if _ops.hip_available():
                 convert_to_fused_bn(model)
+                if cl:
+                    # custom NHWC stem kernels (MIOpen's NHWC bf16 path
+                    # falls to naive_conv for 3-channel stems)
+                    from fedtorch_amd.ops.stemconv import convert_stem
+                    convert_stem(model)
     arena = Arena(model)
     criterion = define_criterion(args)
     if args.graph.on_cuda and torch.cuda.is_available():

@@ -12,6 +12,26 @@ import torch.nn as nn
 
 from fedtorch_amd import ops
 
+_VN = {torch.bfloat16: 8, torch.float32: 4, torch.float64: 2}
+
+
+def _is_cl(x):
+    """channels_last-contiguous 4D (and not plainly contiguous)."""
+    return (x.dim() == 4
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and not x.is_contiguous())
+
+
+def _nhwc_ok(C, dtype):
+    """Eligibility of the NHWC kernel path (ops/hip/batchnorm.h bnh_*):
+    C a multiple of the 16 B vector width with a power-of-two group count
+    ≤ 64 and C ≤ 256."""
+    vn = _VN.get(dtype)
+    if vn is None or C % vn or C > 256:
+        return False
+    cgc = C // vn
+    return cgc <= 64 and (cgc & (cgc - 1)) == 0
+
 
 class _FusedBNFunction(torch.autograd.Function):
     """y = [relu](bn(x) [+ res]); res fuses a ResNet residual add into the
@@ -31,6 +51,7 @@ class _FusedBNFunction(torch.autograd.Function):
             res if res is not None else empty)
         ctx.relu = bool(relu)
         ctx.has_res = res is not None
+        ctx.nhwc = _is_cl(x)
         if relu:
             ctx.save_for_backward(x, weight, save_mean, save_ivar, y)
         else:
@@ -44,9 +65,13 @@ class _FusedBNFunction(torch.autograd.Function):
         else:
             x, weight, save_mean, save_ivar = ctx.saved_tensors
             y = torch.empty(0, device=x.device)
+        if ctx.nhwc:
+            dy = dy.contiguous(memory_format=torch.channels_last)
+        else:
+            dy = dy.contiguous()
         empty = torch.empty(0, device=x.device)
         dx, dweight, dbias, dres = ops._C.bn_bwd(
-            dy.contiguous(), x, y, save_mean, save_ivar,
+            dy, x, y, save_mean, save_ivar,
             weight if weight is not None else empty, ctx.relu, ctx.has_res)
         return (dx, dweight if weight is not None else None,
                 dbias if weight is not None else None, None, None, None,
@@ -62,8 +87,10 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
     _nbt_pending = 0  # lazy num_batches_tracked increments (flushed on save)
 
     def forward(self, x, res=None):
+        cl = _is_cl(x)
         use_fused = (self.training and x.is_cuda and x.dim() == 4
-                     and ops.hip_available() and not ops.FORCE_EAGER)
+                     and ops.hip_available() and not ops.FORCE_EAGER
+                     and (not cl or _nhwc_ok(self.num_features, x.dtype)))
         if not use_fused:
             self._flush_nbt()
             y = super().forward(x)
@@ -77,13 +104,19 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
             # state_dict is read (checkpoint / sync).
             self._nbt_pending += 1
         momentum = self.momentum if self.momentum is not None else 0.1
+        if cl:
+            xc = x  # already channels_last-contiguous
+            rc = (res.contiguous(memory_format=torch.channels_last)
+                  if res is not None else None)
+        else:
+            xc = x.contiguous()
+            rc = res.contiguous() if res is not None else None
         return _FusedBNFunction.apply(
-            x.contiguous(), self.weight, self.bias,
+            xc, self.weight, self.bias,
             self.running_mean if self.track_running_stats else None,
             self.running_var if self.track_running_stats else None,
             momentum, self.eps,
-            self.fuse_relu or res is not None,
-            res.contiguous() if res is not None else None)
+            self.fuse_relu or res is not None, rc)
 
     def _flush_nbt(self):
         if self._nbt_pending and self.num_batches_tracked is not None:

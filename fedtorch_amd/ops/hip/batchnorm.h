@@ -203,6 +203,290 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
   }
 }
 
+// ==========================================================================
+// NHWC (channels_last) variants — the end-to-end NHWC path runs MIOpen's
+// native-NHWC igemm conv solvers WITHOUT the batched_transpose /
+// SubTensorOp wrapper kernels that NCHW tensors force around them
+// (~45 % of ResNet-20 step kernel time at b256, profiles/r01_bench_notes.md).
+//
+// Layout: element (n, i=h*W+w, c) at (n*HW + i)*C + c — channels contiguous.
+// Thread task split: each thread owns one channel GROUP (VN consecutive
+// channels = one 16 B vector) of one row and keeps VN fp32 partials in
+// registers; the cross-row reduction is a wave64 xor-butterfly over lanes
+// with the same channel-group residue (offsets CGC..32, CGC = C/VN, a power
+// of two ≤ 64), then one LDS pass across the 4 waves.  No atomics, no
+// zero-fill: per-block partials land in part[B, C, 2] and the consumer
+// kernel finalizes per channel (same scheme as the NCHW pack above).
+// Host guarantees: C % VN == 0, CGC = C/VN is a power of two ≤ 64, C ≤ 256.
+// ==========================================================================
+
+// lgc = log2(C/VN); rows NI = N*HW.
+template <typename T, typename VT, int VN>
+__global__ void bnh_stats_k(const T* __restrict__ x, long NI, long C,
+                            int lgc, float* __restrict__ part /*[B,C,2]*/) {
+  const int cgc = 1 << lgc;
+  const unsigned t = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned cg = t & (cgc - 1);
+  const unsigned rstride = (gridDim.x * blockDim.x) >> lgc;
+  float s[VN], q[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) { s[j] = 0.f; q[j] = 0.f; }
+  for (unsigned r = t >> lgc; r < NI; r += rstride) {
+    VT v = *reinterpret_cast<const VT*>(x + (long)r * C + cg * VN);
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      float f = (float)v.d[j];
+      s[j] += f;
+      q[j] = fmaf(f, f, q[j]);
+    }
+  }
+  // wave butterfly over same-residue lanes
+  for (int off = WAVE / 2; off >= cgc; off >>= 1) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      s[j] += __shfl_xor(s[j], off, WAVE);
+      q[j] += __shfl_xor(q[j], off, WAVE);
+    }
+  }
+  __shared__ float lds[2][FT_BLOCK / WAVE][256];  // [s|q][wave][channel]
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  if (lane < cgc) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      lds[0][wid][lane * VN + j] = s[j];
+      lds[1][wid][lane * VN + j] = q[j];
+    }
+  }
+  __syncthreads();
+  const int c = threadIdx.x;
+  if (c < C) {
+    float as = 0.f, aq = 0.f;
+#pragma unroll
+    for (int w = 0; w < FT_BLOCK / WAVE; ++w) {
+      as += lds[0][w][c];
+      aq += lds[1][w][c];
+    }
+    float* p = part + ((long)blockIdx.x * C + c) * 2;
+    p[0] = as;
+    p[1] = aq;
+  }
+}
+
+// finalize per-channel scale/shift into LDS, then stream normalize
+// [+res add][+ReLU].  Block 0 threads c<C write saved/running stats.
+template <typename T, typename VT, int VN>
+__global__ void bnh_norm_k(const T* __restrict__ x, T* __restrict__ y,
+                           const float* __restrict__ part, int B,
+                           const float* __restrict__ weight,
+                           const float* __restrict__ bias,
+                           float* __restrict__ save_mean,
+                           float* __restrict__ save_ivar,
+                           float* __restrict__ running_mean,
+                           float* __restrict__ running_var,
+                           const T* __restrict__ res, long NI, long C,
+                           int lgc, float eps, float momentum, int relu) {
+  __shared__ float lsc[256], lsh[256];
+  const float inv_n = 1.0f / (float)NI;
+  const int c = threadIdx.x;
+  if (c < C) {
+    float s = 0.f, q = 0.f;
+    for (int b = 0; b < B; ++b) {
+      const float* p = part + ((long)b * C + c) * 2;
+      s += p[0];
+      q += p[1];
+    }
+    const float mean = s * inv_n;
+    const float var = fmaxf(q * inv_n - mean * mean, 0.f);
+    const float ivar = rsqrtf(var + eps);
+    const float w = weight ? weight[c] : 1.f;
+    const float bb = bias ? bias[c] : 0.f;
+    lsc[c] = w * ivar;
+    lsh[c] = bb - mean * (w * ivar);
+    if (blockIdx.x == 0) {
+      save_mean[c] = mean;
+      save_ivar[c] = ivar;
+      if (running_mean) {
+        float ub = var * (float)NI / (float)(NI > 1 ? NI - 1 : 1);
+        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+        running_var[c] = (1.f - momentum) * running_var[c] + momentum * ub;
+      }
+    }
+  }
+  __syncthreads();
+  const int cgc = 1 << lgc;
+  const unsigned t = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned cg = t & (cgc - 1);
+  const unsigned rstride = (gridDim.x * blockDim.x) >> lgc;
+  float sc[VN], sh[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) {
+    sc[j] = lsc[cg * VN + j];
+    sh[j] = lsh[cg * VN + j];
+  }
+  for (unsigned r = t >> lgc; r < NI; r += rstride) {
+    const long base = (long)r * C + cg * VN;
+    VT v = *reinterpret_cast<const VT*>(x + base);
+    if (res) {
+      VT rv = *reinterpret_cast<const VT*>(res + base);
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float f = fmaf((float)v.d[j], sc[j], sh[j]) + (float)rv.d[j];
+        v.d[j] = (T)(relu ? fmaxf(f, 0.f) : f);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float f = fmaf((float)v.d[j], sc[j], sh[j]);
+        v.d[j] = (T)(relu ? fmaxf(f, 0.f) : f);
+      }
+    }
+    *reinterpret_cast<VT*>(y + base) = v;
+  }
+}
+
+template <typename T, typename VT, int VN>
+__global__ void bnh_bwd_stats_k(const T* __restrict__ dy,
+                                const T* __restrict__ x,
+                                const T* __restrict__ yv,
+                                const float* __restrict__ save_mean,
+                                const float* __restrict__ save_ivar, long NI,
+                                long C, int lgc,
+                                float* __restrict__ part /*[B,C,2]*/,
+                                int relu) {
+  const int cgc = 1 << lgc;
+  const unsigned t = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned cg = t & (cgc - 1);
+  const unsigned rstride = (gridDim.x * blockDim.x) >> lgc;
+  float mean[VN], ivar[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) {
+    mean[j] = save_mean[cg * VN + j];
+    ivar[j] = save_ivar[cg * VN + j];
+  }
+  float s1[VN], s2[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) { s1[j] = 0.f; s2[j] = 0.f; }
+  for (unsigned r = t >> lgc; r < NI; r += rstride) {
+    const long base = (long)r * C + cg * VN;
+    VT g = *reinterpret_cast<const VT*>(dy + base);
+    VT xv = *reinterpret_cast<const VT*>(x + base);
+    if (relu) {
+      VT yy = *reinterpret_cast<const VT*>(yv + base);
+#pragma unroll
+      for (int j = 0; j < VN; ++j)
+        if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      float gg = (float)g.d[j];
+      float xh = ((float)xv.d[j] - mean[j]) * ivar[j];
+      s1[j] += gg;
+      s2[j] = fmaf(gg, xh, s2[j]);
+    }
+  }
+  for (int off = WAVE / 2; off >= cgc; off >>= 1) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      s1[j] += __shfl_xor(s1[j], off, WAVE);
+      s2[j] += __shfl_xor(s2[j], off, WAVE);
+    }
+  }
+  __shared__ float lds[2][FT_BLOCK / WAVE][256];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  if (lane < cgc) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      lds[0][wid][lane * VN + j] = s1[j];
+      lds[1][wid][lane * VN + j] = s2[j];
+    }
+  }
+  __syncthreads();
+  const int c = threadIdx.x;
+  if (c < C) {
+    float a1 = 0.f, a2 = 0.f;
+#pragma unroll
+    for (int w = 0; w < FT_BLOCK / WAVE; ++w) {
+      a1 += lds[0][w][c];
+      a2 += lds[1][w][c];
+    }
+    float* p = part + ((long)blockIdx.x * C + c) * 2;
+    p[0] = a1;
+    p[1] = a2;
+  }
+}
+
+template <typename T, typename VT, int VN>
+__global__ void bnh_bwd_dx_k(const T* __restrict__ dy,
+                             const T* __restrict__ x,
+                             const T* __restrict__ yv,
+                             const float* __restrict__ part, int B,
+                             const float* __restrict__ save_mean,
+                             const float* __restrict__ save_ivar,
+                             const float* __restrict__ weight,
+                             T* __restrict__ dx, T* __restrict__ dres,
+                             float* __restrict__ dweight,
+                             float* __restrict__ dbias, long NI, long C,
+                             int lgc, int relu) {
+  __shared__ float lk1[256], lmdy[256], lmdyxh[256], lmean[256], livar[256];
+  const float inv_n = 1.0f / (float)NI;
+  const int c = threadIdx.x;
+  if (c < C) {
+    float s1 = 0.f, s2 = 0.f;
+    for (int b = 0; b < B; ++b) {
+      const float* p = part + ((long)b * C + c) * 2;
+      s1 += p[0];
+      s2 += p[1];
+    }
+    const float w = weight ? weight[c] : 1.f;
+    const float iv = save_ivar[c];
+    lk1[c] = w * iv;
+    lmdy[c] = s1 * inv_n;
+    lmdyxh[c] = s2 * inv_n;
+    lmean[c] = save_mean[c];
+    livar[c] = iv;
+    if (blockIdx.x == 0) {
+      dweight[c] = s2;
+      dbias[c] = s1;
+    }
+  }
+  __syncthreads();
+  const int cgc = 1 << lgc;
+  const unsigned t = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned cg = t & (cgc - 1);
+  const unsigned rstride = (gridDim.x * blockDim.x) >> lgc;
+  float k1[VN], mdy[VN], mdyxh[VN], mean[VN], ivar[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) {
+    const int cc = cg * VN + j;
+    k1[j] = lk1[cc];
+    mdy[j] = lmdy[cc];
+    mdyxh[j] = lmdyxh[cc];
+    mean[j] = lmean[cc];
+    ivar[j] = livar[cc];
+  }
+  for (unsigned r = t >> lgc; r < NI; r += rstride) {
+    const long base = (long)r * C + cg * VN;
+    VT g = *reinterpret_cast<const VT*>(dy + base);
+    VT xv = *reinterpret_cast<const VT*>(x + base);
+    if (relu) {
+      VT yy = *reinterpret_cast<const VT*>(yv + base);
+#pragma unroll
+      for (int j = 0; j < VN; ++j)
+        if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+    }
+    if (dres) *reinterpret_cast<VT*>(dres + base) = g;
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      float gg = (float)g.d[j];
+      float xh = ((float)xv.d[j] - mean[j]) * ivar[j];
+      g.d[j] = (T)(k1[j] * (gg - mdy[j] - xh * mdyxh[j]));
+    }
+    *reinterpret_cast<VT*>(dx + base) = g;
+  }
+}
+
 // dx = (gamma*ivar) * (dy - sum(dy)/n - xhat * sum(dy*xhat)/n)
 // block (0, c) lane 0 writes dgamma[c] = sum(dy*xhat), dbeta[c] = sum(dy).
 // `dres` (optional): gradient of the fused residual input = ReLU-masked dy.

@@ -13,6 +13,7 @@
 
 #include "common.h"
 #include "batchnorm.h"
+#include "stemconv.h"
 
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 #define STREAM at::hip::getCurrentHIPStream().stream()
@@ -739,6 +740,87 @@ void gather_grads(torch::Tensor srcs, torch::Tensor table,
 }
 
 // ==========================================================================
+// NHWC stem convolution (kernels in stemconv.h)
+// ==========================================================================
+static inline bool stem_cl(const torch::Tensor& t) {
+  return t.is_contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, bool out_bf16) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && stem_cl(x),
+              "stem_conv_fwd: x must be 4D channels_last on GPU");
+  TORCH_CHECK(w.is_cuda() && w.dim() == 4 && stem_cl(w)
+                  && w.scalar_type() == torch::kFloat,
+              "stem_conv_fwd: w must be fp32 channels_last");
+  const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = w.size(0);
+  TORCH_CHECK(w.size(1) == Ci && w.size(2) == 3 && w.size(3) == 3
+                  && Ci <= STEM_MAX_CI && Co <= STEM_MAX_CO && 256 % Co == 0,
+              "stem_conv_fwd: unsupported shape");
+  auto y = torch::empty({N, Co, H, W},
+                        x.options()
+                            .dtype(out_bf16 ? torch::kBFloat16
+                                            : torch::kFloat)
+                            .memory_format(at::MemoryFormat::ChannelsLast));
+  const long total = (long)N * H * W;
+  const int grid = ft_grid(total);
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "stem_fwd", [&] {
+    using TX = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                  __hip_bfloat16, scalar_t>;
+    if (out_bf16)
+      hipLaunchKernelGGL((stem_fwd_k<TX, __hip_bfloat16>), dim3(grid),
+                         dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const TX*>(x.data_ptr()),
+                         w.data_ptr<float>(),
+                         reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                         N, H, W, Ci, Co);
+    else
+      hipLaunchKernelGGL((stem_fwd_k<TX, float>), dim3(grid),
+                         dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const TX*>(x.data_ptr()),
+                         w.data_ptr<float>(),
+                         reinterpret_cast<float*>(y.data_ptr()),
+                         N, H, W, Ci, Co);
+  });
+  return y;
+}
+
+torch::Tensor stem_conv_wrw(torch::Tensor dy, torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && stem_cl(x), "stem_wrw: x");
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && stem_cl(dy), "stem_wrw: dy");
+  const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = dy.size(1);
+  TORCH_CHECK(Ci <= STEM_MAX_CI && Co <= STEM_MAX_CO && 256 % Co == 0,
+              "stem_wrw: unsupported shape");
+  const int wn = Co * 9 * Ci;
+  const int B = 64;
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto part = torch::empty({B, wn}, f32);
+  auto dw = torch::empty(
+      {Co, Ci, 3, 3}, f32.memory_format(at::MemoryFormat::ChannelsLast));
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "stem_wrw_x", [&] {
+    using TX = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                  __hip_bfloat16, scalar_t>;
+    AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16,
+                                   dy.scalar_type(), "stem_wrw_dy", [&] {
+      using TY = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                    __hip_bfloat16, scalar_t>;
+      hipLaunchKernelGGL((stem_wrw_k<TY, TX>), dim3(B), dim3(FT_BLOCK), 0,
+                         STREAM,
+                         reinterpret_cast<const TY*>(dy.data_ptr()),
+                         reinterpret_cast<const TX*>(x.data_ptr()),
+                         part.data_ptr<float>(), N, H, W, Ci, Co);
+    });
+  });
+  hipLaunchKernelGGL(stem_wrw_final_k, dim3((wn + FT_BLOCK - 1) / FT_BLOCK),
+                     dim3(FT_BLOCK), 0, STREAM, part.data_ptr<float>(), B,
+                     wn, dw.data_ptr<float>());
+  return dw;
+}
+
+// ==========================================================================
 // fused spatial BatchNorm (training fwd/bwd) — kernels in batchnorm.h
 // ==========================================================================
 // number of per-channel partial blocks for the reduction kernels: target
@@ -762,14 +844,39 @@ static inline int bn_ew_chunks(long per_v, long C) {
   return (int)(c < 1 ? 1 : c);
 }
 
+// NHWC eligibility: C a multiple of the 16 B vector width with a
+// power-of-two group count ≤ 64 lanes and ≤ 256 (blockDim) channels.
+static inline bool bnh_ok(long C, int VN) {
+  if (C <= 0 || C > 256 || C % VN) return false;
+  long cgc = C / VN;
+  return cgc <= 64 && (cgc & (cgc - 1)) == 0;
+}
+static inline int bnh_lgc(long C, int VN) {
+  int l = 0;
+  for (long c = C / VN; c > 1; c >>= 1) ++l;
+  return l;
+}
+static inline int bnh_red_grid(long tasks) {
+  long b = (tasks + FT_BLOCK * 8 - 1) / (FT_BLOCK * 8);
+  if (b > 64) b = 64;
+  return (int)(b < 1 ? 1 : b);
+}
+static inline int bnh_ew_grid(long tasks) {
+  long b = (tasks + FT_BLOCK * 2 - 1) / (FT_BLOCK * 2);
+  if (b > 1024) b = 1024;
+  return (int)(b < 1 ? 1 : b);
+}
+
 std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor bias,
                                         torch::Tensor running_mean,
                                         torch::Tensor running_var,
                                         double eps, double momentum,
                                         bool relu, torch::Tensor res) {
-  CHK(x);
-  TORCH_CHECK(x.dim() == 4, "bn_fwd_train expects NCHW");
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn_fwd_train expects 4D GPU");
+  const bool nhwc =
+      x.is_contiguous(at::MemoryFormat::ChannelsLast) && !x.is_contiguous();
+  TORCH_CHECK(nhwc || x.is_contiguous(), "bn_fwd_train: x not contiguous");
   long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   TORCH_CHECK(N * HW < (1L << 31), "bn_fwd_train: N*HW must fit in int32");
   auto f32 = x.options().dtype(torch::kFloat);
@@ -777,6 +884,40 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
   auto save_ivar = torch::empty({C}, f32);
   auto y = torch::empty_like(x);
   bool track = running_mean.numel() > 0;
+  if (nhwc) {
+    AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                   "bnh_fwd", [&] {
+      using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                   __hip_bfloat16, scalar_t>;
+      constexpr int VN = BnVec<T>::N;
+      TORCH_CHECK(bnh_ok(C, VN), "bn_fwd_train NHWC: unsupported C=", C);
+      const int lgc = bnh_lgc(C, VN);
+      const long NI = N * HW, tasks = NI << lgc;
+      const int B = bnh_red_grid(tasks);
+      auto part = torch::empty({B, C, 2}, f32);
+      hipLaunchKernelGGL((bnh_stats_k<T, typename BnVec<T>::V, VN>),
+                         dim3(B), dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const T*>(x.data_ptr()), NI, C,
+                         lgc, part.data_ptr<float>());
+      hipLaunchKernelGGL((bnh_norm_k<T, typename BnVec<T>::V, VN>),
+                         dim3(bnh_ew_grid(tasks)), dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const T*>(x.data_ptr()),
+                         reinterpret_cast<T*>(y.data_ptr()),
+                         part.data_ptr<float>(), B,
+                         weight.numel() ? weight.data_ptr<float>() : nullptr,
+                         bias.numel() ? bias.data_ptr<float>() : nullptr,
+                         save_mean.data_ptr<float>(),
+                         save_ivar.data_ptr<float>(),
+                         track ? running_mean.data_ptr<float>() : nullptr,
+                         track ? running_var.data_ptr<float>() : nullptr,
+                         res.numel()
+                             ? reinterpret_cast<const T*>(res.data_ptr())
+                             : nullptr,
+                         NI, C, lgc, (float)eps, (float)momentum,
+                         relu ? 1 : 0);
+    });
+    return {y, save_mean, save_ivar};
+  }
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
                                  "bn_fwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
@@ -816,7 +957,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor save_ivar,
                                   torch::Tensor weight, bool relu,
                                   bool has_res) {
-  CHK(dy); CHK(x);
+  TORCH_CHECK(x.is_cuda() && dy.is_cuda(), "bn_bwd expects GPU tensors");
+  const bool nhwc =
+      x.is_contiguous(at::MemoryFormat::ChannelsLast) && !x.is_contiguous();
   long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
   TORCH_CHECK(N * HW < (1L << 31), "bn_bwd: N*HW must fit in int32");
   auto f32 = x.options().dtype(torch::kFloat);
@@ -825,6 +968,45 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                       : torch::empty({0}, x.options());
   auto dweight = torch::empty({C}, f32);
   auto dbias = torch::empty({C}, f32);
+  if (nhwc) {
+    TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "bn_bwd NHWC: dy must be channels_last");
+    AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                   "bnh_bwd", [&] {
+      using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                   __hip_bfloat16, scalar_t>;
+      constexpr int VN = BnVec<T>::N;
+      TORCH_CHECK(bnh_ok(C, VN), "bn_bwd NHWC: unsupported C=", C);
+      const int lgc = bnh_lgc(C, VN);
+      const long NI = N * HW, tasks = NI << lgc;
+      const int B = bnh_red_grid(tasks);
+      auto part = torch::empty({B, C, 2}, f32);
+      const T* yp = relu ? reinterpret_cast<const T*>(y.data_ptr()) : nullptr;
+      hipLaunchKernelGGL((bnh_bwd_stats_k<T, typename BnVec<T>::V, VN>),
+                         dim3(B), dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const T*>(dy.data_ptr()),
+                         reinterpret_cast<const T*>(x.data_ptr()), yp,
+                         save_mean.data_ptr<float>(),
+                         save_ivar.data_ptr<float>(), NI, C, lgc,
+                         part.data_ptr<float>(), relu ? 1 : 0);
+      hipLaunchKernelGGL((bnh_bwd_dx_k<T, typename BnVec<T>::V, VN>),
+                         dim3(bnh_ew_grid(tasks)), dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const T*>(dy.data_ptr()),
+                         reinterpret_cast<const T*>(x.data_ptr()), yp,
+                         part.data_ptr<float>(), B,
+                         save_mean.data_ptr<float>(),
+                         save_ivar.data_ptr<float>(),
+                         weight.numel() ? weight.data_ptr<float>() : nullptr,
+                         reinterpret_cast<T*>(dx.data_ptr()),
+                         has_res ? reinterpret_cast<T*>(dres.data_ptr())
+                                 : nullptr,
+                         dweight.data_ptr<float>(), dbias.data_ptr<float>(),
+                         NI, C, lgc, relu ? 1 : 0);
+    });
+    return {dx, dweight, dbias, dres};
+  }
+  TORCH_CHECK(x.is_contiguous() && dy.is_contiguous(),
+              "bn_bwd: x/dy must be contiguous");
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
                                  "bn_bwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
@@ -880,4 +1062,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scatter_accumulate", &scatter_accumulate);
   m.def("multi_diff_accumulate", &multi_diff_accumulate);
   m.def("gather_grads", &gather_grads);
+  m.def("stem_conv_fwd", &stem_conv_fwd);
+  m.def("stem_conv_wrw", &stem_conv_wrw);
 }

@@ -53,11 +53,20 @@ class Arena(object):
         self.dtype = dtype
         self.device = device if device is not None else params[0].device
         self.offsets, self.numels, self.shapes = [], [], []
+        # channels_last 4D params are packed in NHWC element order so the
+        # arena view handed back to the module keeps channels_last strides
+        # (the end-to-end NHWC path needs conv weights in that format; every
+        # arena op is elementwise over the flat buffer, so order is free).
+        self.cl = []
         total = 0
         for p in params:
             self.offsets.append(total)
             self.numels.append(p.numel())
             self.shapes.append(p.shape)
+            self.cl.append(
+                p.dim() == 4
+                and p.data.is_contiguous(memory_format=torch.channels_last)
+                and not p.data.is_contiguous())
             total += _aligned(p.numel())
         # wd applies to flat[:wd_numel] (includes pad gaps, which stay zero).
         self.wd_numel = (self.offsets[len(decay) - 1] +
@@ -69,9 +78,14 @@ class Arena(object):
         # NOTE: the pad gaps stay zero forever: every fused op is linear in
         # the buffers, so zero gaps stay zero through steps/collectives.
         self.flat = torch.zeros(total, dtype=dtype, device=self.device)
-        for p, off, n in zip(params, self.offsets, self.numels):
-            self.flat[off:off + n].copy_(p.data.reshape(-1).to(dtype))
-            p.data = self.flat[off:off + n].view(p.shape)
+        for p, off, n, cl in zip(params, self.offsets, self.numels, self.cl):
+            if cl:
+                nhwc = p.data.permute(0, 2, 3, 1)
+                self.flat[off:off + n].copy_(nhwc.reshape(-1).to(dtype))
+                p.data = self._cl_view(self.flat, off, n, p.shape)
+            else:
+                self.flat[off:off + n].copy_(p.data.reshape(-1).to(dtype))
+                p.data = self.flat[off:off + n].view(p.shape)
         self.params = params
         self.grad = None
         if with_grads:
@@ -102,11 +116,21 @@ class Arena(object):
             self.buf_flat[off:off + n].copy_(m._buffers[bname].reshape(-1))
             m._buffers[bname] = self.buf_flat[off:off + n]
 
+    @staticmethod
+    def _cl_view(flat, off, n, shape):
+        """channels_last view of an arena region storing NHWC element order."""
+        N, C, H, W = shape
+        return flat[off:off + n].view(N, H, W, C).permute(0, 3, 1, 2)
+
     # ---- gradient plumbing -------------------------------------------------
     def attach_grads(self):
         """Point every p.grad at its view of the grad arena."""
-        for p, off, n in zip(self.params, self.offsets, self.numels):
-            p.grad = self.grad[off:off + n].view(p.shape)
+        for p, off, n, cl in zip(self.params, self.offsets, self.numels,
+                                 self.cl):
+            if cl:
+                p.grad = self._cl_view(self.grad, off, n, p.shape)
+            else:
+                p.grad = self.grad[off:off + n].view(p.shape)
         self._gather_state = None
 
     def zero_grad(self):
@@ -127,11 +151,16 @@ class Arena(object):
 
     def _build_gather_state(self):
         srcs, table = [], []
-        for i, (p, off, n) in enumerate(zip(self.params, self.offsets,
-                                            self.numels)):
+        for i, (p, off, n, cl) in enumerate(zip(self.params, self.offsets,
+                                                self.numels, self.cl)):
             g = p.grad
-            if (g is None or not g.is_contiguous()
-                    or g.dtype != torch.float32):
+            # a channels_last stolen grad's linear element order matches the
+            # NHWC-packed arena region, so its flat copy is still direct
+            dense = (g is not None
+                     and (g.is_contiguous(
+                             memory_format=torch.channels_last)
+                          if cl else g.is_contiguous()))
+            if not dense or g.dtype != torch.float32:
                 return None  # fall back to per-tensor copies
             srcs.append(g.data_ptr())
             for c in range(0, n, self._GATHER_CHUNK):
@@ -152,8 +181,10 @@ class Arena(object):
             ops._C.gather_grads(self._gather_state[0],
                                 self._gather_state[1], self.grad)
         else:  # non-contiguous / non-fp32 stolen grads: plain copies
-            for p, off, n in zip(self.params, self.offsets, self.numels):
-                self.grad[off:off + n].copy_(p.grad.reshape(-1))
+            for p, off, n, cl in zip(self.params, self.offsets, self.numels,
+                                     self.cl):
+                g = p.grad.permute(0, 2, 3, 1) if cl else p.grad
+                self.grad[off:off + n].copy_(g.reshape(-1))
 
     # ---- flat state helpers ------------------------------------------------
     def new_buffer(self, zero=True):

@@ -357,6 +357,127 @@ def test_gather_grads_matches_attached():
     assert torch.allclose(a2.grad, ref, atol=1e-6)
 
 
+@pytest.mark.parametrize('C', [16, 32, 64])
+@pytest.mark.parametrize('mode', ['plain', 'relu', 'addrelu'])
+def test_fused_bn_nhwc_matches_torch(C, mode):
+    """NHWC (channels_last) BN kernels vs eager reference."""
+    import torch.nn as nn
+    from fedtorch_amd.ops.batchnorm import FusedBatchNorm2d, BNAddReLU, \
+        convert_to_fused_bn
+    torch.manual_seed(31)
+    N, H, W = 8, 10, 10
+    cl = torch.channels_last
+    x = torch.randn(N, C, H, W, device='cuda').contiguous(memory_format=cl)
+    ref = nn.BatchNorm2d(C).cuda()
+    ref.weight.data.uniform_(0.5, 1.5)
+    ref.bias.data.uniform_(-0.5, 0.5)
+    xr = x.clone().requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    if mode == 'addrelu':
+        res = torch.randn_like(x).contiguous(memory_format=cl)
+        rr = res.clone().requires_grad_(True)
+        rf = res.clone().requires_grad_(True)
+        fus = BNAddReLU(C).cuda()
+        fus.bn.load_state_dict(ref.state_dict())
+        convert_to_fused_bn(fus)
+        yr = torch.relu(ref(xr) + rr)
+        yf = fus(xf, rf)
+        fw, fb = fus.bn.weight, fus.bn.bias
+        frm, frv = fus.bn.running_mean, fus.bn.running_var
+    else:
+        fus = FusedBatchNorm2d(C).cuda()
+        fus.fuse_relu = mode == 'relu'
+        fus.load_state_dict(ref.state_dict())
+        yr = ref(xr)
+        if mode == 'relu':
+            yr = torch.relu(yr)
+        yf = fus(xf)
+        fw, fb = fus.weight, fus.bias
+        frm, frv = fus.running_mean, fus.running_var
+    assert yf.is_contiguous(memory_format=cl)
+    assert torch.allclose(yf, yr, atol=3e-5), (yf - yr).abs().max().item()
+    g = torch.randn_like(yr)
+    yr.backward(g)
+    yf.backward(g)
+    assert torch.allclose(xf.grad, xr.grad, atol=3e-4)
+    if mode == 'addrelu':
+        assert torch.allclose(rf.grad, rr.grad, atol=3e-5)
+    assert torch.allclose(fw.grad, ref.weight.grad, atol=2e-3)
+    assert torch.allclose(fb.grad, ref.bias.grad, atol=2e-3)
+    assert torch.allclose(frm, ref.running_mean, atol=1e-5)
+    assert torch.allclose(frv, ref.running_var, atol=1e-4)
+
+
+@pytest.mark.parametrize('xdtype', ['float32', 'bfloat16'])
+def test_nhwc_stem_conv_matches_torch(xdtype):
+    """Custom NHWC stem conv (fwd + wrw) vs F.conv2d."""
+    import torch.nn.functional as F
+    dt = getattr(torch, xdtype)
+    torch.manual_seed(32)
+    cl = torch.channels_last
+    N, Ci, H, W, Co = 8, 3, 32, 32, 16
+    x = torch.randn(N, Ci, H, W, device='cuda', dtype=dt).contiguous(
+        memory_format=cl)
+    w = torch.randn(Co, Ci, 3, 3, device='cuda').contiguous(
+        memory_format=cl) * 0.1
+    y = ops._C.stem_conv_fwd(x, w, False)
+    ref = F.conv2d(x.float(), w, padding=1)
+    tol = 1e-4 if xdtype == 'float32' else 0.05
+    assert y.is_contiguous(memory_format=cl)
+    assert torch.allclose(y, ref, atol=tol), (y - ref).abs().max().item()
+
+    dy = torch.randn(N, Co, H, W, device='cuda', dtype=dt).contiguous(
+        memory_format=cl)
+    dw = ops._C.stem_conv_wrw(dy, x)
+    wr = w.clone().requires_grad_(True)
+    F.conv2d(x.float(), wr, padding=1).backward(dy.float())
+    wtol = 2e-3 if xdtype == 'float32' else 2.0
+    assert torch.allclose(dw, wr.grad, atol=wtol, rtol=1e-2), \
+        (dw - wr.grad).abs().max().item()
+
+
+def test_nhwc_resnet20_step_matches_nchw():
+    """Full resnet20: one bf16 fwd/bwd in channels_last (fused NHWC BN +
+    custom stem) tracks the NCHW eager path."""
+    from types import SimpleNamespace
+    from fedtorch_amd.components.models.resnet import resnet
+    from fedtorch_amd.ops.batchnorm import convert_to_fused_bn
+    from fedtorch_amd.ops.stemconv import convert_stem, NhwcStemConv
+    from fedtorch_amd.parallel.arena import Arena
+    a = SimpleNamespace(arch='resnet20', data='cifar10')
+    torch.manual_seed(33)
+    x = torch.randn(16, 3, 32, 32, device='cuda')
+    y = torch.randint(0, 10, (16,), device='cuda')
+    crit = torch.nn.CrossEntropyLoss()
+
+    def build(cl):
+        torch.manual_seed(5)
+        m = resnet(a).cuda()
+        if cl:
+            m = m.to(memory_format=torch.channels_last)
+            convert_to_fused_bn(m)
+            convert_stem(m)
+        return m, Arena(m)
+
+    m_ref, a_ref = build(False)
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        loss_ref = crit(m_ref(x), y)
+    loss_ref.backward()
+
+    m_cl, a_cl = build(True)
+    assert isinstance(m_cl.conv1, NhwcStemConv)
+    xc = x.contiguous(memory_format=torch.channels_last)
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        loss_cl = crit(m_cl(xc), y)
+    loss_cl.backward()
+    assert abs(loss_cl.item() - loss_ref.item()) < 0.05
+    # grad arenas carry the same energy (bf16 noise aside); element order
+    # differs (NHWC packing) so compare norms
+    n_ref = a_ref.grad.norm().item()
+    n_cl = a_cl.grad.norm().item()
+    assert abs(n_cl - n_ref) / max(n_ref, 1e-6) < 0.1, (n_ref, n_cl)
+
+
 def test_fused_bn_nbt_lazy_flush():
     """num_batches_tracked counts on the host and flushes into the buffer
     when the state_dict is read (no per-step GPU kernel)."""
