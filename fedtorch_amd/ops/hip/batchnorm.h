@@ -46,15 +46,24 @@ template <typename T> struct Bn1 {
   static constexpr int N = 1;
 };
 
-// sum the B partials of channel c (uniform across the block; B ≤ 64)
+// sum the B (≤64) partials of channel c: lane-parallel loads + wave64
+// xor-butterfly, so the ~300-cycle L2 latencies overlap instead of forming
+// a B-deep dependent chain (a serial loop here measured ~10 us per call).
+// Every lane of every wave ends up with the full sums.
 __device__ __forceinline__ void bn_finalize(const float* __restrict__ part,
                                             long c, int B, float* s,
                                             float* q) {
+  const int lane = threadIdx.x & (WAVE - 1);
   float a = 0.f, b = 0.f;
   const float* p = part + c * (long)B * 2;
-  for (int i = 0; i < B; ++i) {
+  for (int i = lane; i < B; i += WAVE) {
     a += p[2 * i];
     b += p[2 * i + 1];
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off; off >>= 1) {
+    a += __shfl_xor(a, off, WAVE);
+    b += __shfl_xor(b, off, WAVE);
   }
   *s = a;
   *q = b;
@@ -286,15 +295,31 @@ __global__ void bnh_norm_k(const T* __restrict__ x, T* __restrict__ y,
                            float* __restrict__ running_var,
                            const T* __restrict__ res, long NI, long C,
                            int lgc, float eps, float momentum, int relu) {
-  __shared__ float lsc[256], lsh[256];
+  // block-parallel finalize: thread (p, c) sums a slice of the B partials
+  // so the L2 latencies overlap (a per-channel serial loop here is a
+  // B-deep dependent chain, ~10 us); LDS tree collapses the P slices.
+  __shared__ float lsc[256], lsh[256], fa[FT_BLOCK], fb[FT_BLOCK];
   const float inv_n = 1.0f / (float)NI;
+  {
+    const int P = blockDim.x / (int)C;
+    const int c0 = threadIdx.x % (int)C, p0 = threadIdx.x / (int)C;
+    float a = 0.f, bb = 0.f;
+    for (int i = p0; i < B; i += P) {
+      const float* p = part + ((long)i * C + c0) * 2;
+      a += p[0];
+      bb += p[1];
+    }
+    fa[threadIdx.x] = a;
+    fb[threadIdx.x] = bb;
+  }
+  __syncthreads();
   const int c = threadIdx.x;
   if (c < C) {
     float s = 0.f, q = 0.f;
-    for (int b = 0; b < B; ++b) {
-      const float* p = part + ((long)b * C + c) * 2;
-      s += p[0];
-      q += p[1];
+    const int P = blockDim.x / (int)C;
+    for (int i = 0; i < P; ++i) {
+      s += fa[i * C + c];
+      q += fb[i * C + c];
     }
     const float mean = s * inv_n;
     const float var = fmaxf(q * inv_n - mean * mean, 0.f);
@@ -430,14 +455,28 @@ __global__ void bnh_bwd_dx_k(const T* __restrict__ dy,
                              float* __restrict__ dbias, long NI, long C,
                              int lgc, int relu) {
   __shared__ float lk1[256], lmdy[256], lmdyxh[256], lmean[256], livar[256];
+  __shared__ float fa[FT_BLOCK], fb[FT_BLOCK];
   const float inv_n = 1.0f / (float)NI;
+  {
+    const int P = blockDim.x / (int)C;
+    const int c0 = threadIdx.x % (int)C, p0 = threadIdx.x / (int)C;
+    float a = 0.f, bb = 0.f;
+    for (int i = p0; i < B; i += P) {
+      const float* p = part + ((long)i * C + c0) * 2;
+      a += p[0];
+      bb += p[1];
+    }
+    fa[threadIdx.x] = a;
+    fb[threadIdx.x] = bb;
+  }
+  __syncthreads();
   const int c = threadIdx.x;
   if (c < C) {
     float s1 = 0.f, s2 = 0.f;
-    for (int b = 0; b < B; ++b) {
-      const float* p = part + ((long)b * C + c) * 2;
-      s1 += p[0];
-      s2 += p[1];
+    const int P = blockDim.x / (int)C;
+    for (int i = 0; i < P; ++i) {
+      s1 += fa[i * C + c];
+      s2 += fb[i * C + c];
     }
     const float w = weight ? weight[c] : 1.f;
     const float iv = save_ivar[c];

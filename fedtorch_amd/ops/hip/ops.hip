@@ -755,8 +755,8 @@ torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, bool out_bf16) {
   const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
   const int Co = w.size(0);
   TORCH_CHECK(w.size(1) == Ci && w.size(2) == 3 && w.size(3) == 3
-                  && Ci <= STEM_MAX_CI && Co <= STEM_MAX_CO && 256 % Co == 0,
-              "stem_conv_fwd: unsupported shape");
+                  && Ci == 3 && (Co == 16 || Co == 32),
+              "stem_conv_fwd: supported shapes are Ci=3, Co in {16,32}");
   auto y = torch::empty({N, Co, H, W},
                         x.options()
                             .dtype(out_bf16 ? torch::kBFloat16
@@ -768,20 +768,31 @@ torch::Tensor stem_conv_fwd(torch::Tensor x, torch::Tensor w, bool out_bf16) {
                                  "stem_fwd", [&] {
     using TX = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                   __hip_bfloat16, scalar_t>;
-    if (out_bf16)
-      hipLaunchKernelGGL((stem_fwd_k<TX, __hip_bfloat16>), dim3(grid),
-                         dim3(FT_BLOCK), 0, STREAM,
-                         reinterpret_cast<const TX*>(x.data_ptr()),
-                         w.data_ptr<float>(),
-                         reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
-                         N, H, W, Ci, Co);
-    else
-      hipLaunchKernelGGL((stem_fwd_k<TX, float>), dim3(grid),
-                         dim3(FT_BLOCK), 0, STREAM,
-                         reinterpret_cast<const TX*>(x.data_ptr()),
-                         w.data_ptr<float>(),
-                         reinterpret_cast<float*>(y.data_ptr()),
-                         N, H, W, Ci, Co);
+    if (out_bf16) {
+      auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
+      if (Co == 16)
+        hipLaunchKernelGGL((stem_fwd_k<TX, __hip_bfloat16, 3, 16>),
+                           dim3(grid), dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           w.data_ptr<float>(), yp, N, H, W);
+      else
+        hipLaunchKernelGGL((stem_fwd_k<TX, __hip_bfloat16, 3, 32>),
+                           dim3(grid), dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           w.data_ptr<float>(), yp, N, H, W);
+    } else {
+      auto* yp = reinterpret_cast<float*>(y.data_ptr());
+      if (Co == 16)
+        hipLaunchKernelGGL((stem_fwd_k<TX, float, 3, 16>), dim3(grid),
+                           dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           w.data_ptr<float>(), yp, N, H, W);
+      else
+        hipLaunchKernelGGL((stem_fwd_k<TX, float, 3, 32>), dim3(grid),
+                           dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           w.data_ptr<float>(), yp, N, H, W);
+    }
   });
   return y;
 }
@@ -791,10 +802,10 @@ torch::Tensor stem_conv_wrw(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(dy.is_cuda() && dy.dim() == 4 && stem_cl(dy), "stem_wrw: dy");
   const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
   const int Co = dy.size(1);
-  TORCH_CHECK(Ci <= STEM_MAX_CI && Co <= STEM_MAX_CO && 256 % Co == 0,
-              "stem_wrw: unsupported shape");
+  TORCH_CHECK(Ci == 3 && (Co == 16 || Co == 32),
+              "stem_wrw: supported shapes are Ci=3, Co in {16,32}");
   const int wn = Co * 9 * Ci;
-  const int B = 64;
+  const int B = 128;
   auto f32 = x.options().dtype(torch::kFloat);
   auto part = torch::empty({B, wn}, f32);
   auto dw = torch::empty(
@@ -807,11 +818,18 @@ torch::Tensor stem_conv_wrw(torch::Tensor dy, torch::Tensor x) {
                                    dy.scalar_type(), "stem_wrw_dy", [&] {
       using TY = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
                                     __hip_bfloat16, scalar_t>;
-      hipLaunchKernelGGL((stem_wrw_k<TY, TX>), dim3(B), dim3(FT_BLOCK), 0,
-                         STREAM,
-                         reinterpret_cast<const TY*>(dy.data_ptr()),
-                         reinterpret_cast<const TX*>(x.data_ptr()),
-                         part.data_ptr<float>(), N, H, W, Ci, Co);
+      if (Co == 16)
+        hipLaunchKernelGGL((stem_wrw_k<TY, TX, 3, 16>), dim3(B),
+                           dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TY*>(dy.data_ptr()),
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           part.data_ptr<float>(), N, H, W);
+      else
+        hipLaunchKernelGGL((stem_wrw_k<TY, TX, 3, 32>), dim3(B),
+                           dim3(FT_BLOCK), 0, STREAM,
+                           reinterpret_cast<const TY*>(dy.data_ptr()),
+                           reinterpret_cast<const TX*>(x.data_ptr()),
+                           part.data_ptr<float>(), N, H, W);
     });
   });
   hipLaunchKernelGGL(stem_wrw_final_k, dim3((wn + FT_BLOCK - 1) / FT_BLOCK),
@@ -858,7 +876,7 @@ static inline int bnh_lgc(long C, int VN) {
 }
 static inline int bnh_red_grid(long tasks) {
   long b = (tasks + FT_BLOCK * 8 - 1) / (FT_BLOCK * 8);
-  if (b > 64) b = 64;
+  if (b > 128) b = 128;
   return (int)(b < 1 ? 1 : b);
 }
 static inline int bnh_ew_grid(long tasks) {

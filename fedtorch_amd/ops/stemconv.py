@@ -47,8 +47,8 @@ class NhwcStemConv(nn.Conv2d):
                and self.weight.is_contiguous(
                    memory_format=torch.channels_last)
                and self.weight.dtype == torch.float32
-               and self.in_channels <= 4 and self.out_channels <= 32
-               and 256 % self.out_channels == 0
+               and self.in_channels == 3
+               and self.out_channels in (16, 32)
                and ops.hip_available() and not ops.FORCE_EAGER)
         if not use:
             return F.conv2d(x, self.weight, self.bias, self.stride,
@@ -66,7 +66,7 @@ def convert_stem(model):
     if (isinstance(conv, nn.Conv2d) and type(conv) is nn.Conv2d
             and conv.kernel_size == (3, 3) and conv.stride == (1, 1)
             and conv.padding == (1, 1) and conv.bias is None
-            and conv.in_channels <= 4 and conv.out_channels <= 32):
+            and conv.in_channels == 3 and conv.out_channels in (16, 32)):
         new = NhwcStemConv(conv.in_channels, conv.out_channels, 3,
                            stride=1, padding=1, bias=False)
         new.weight = conv.weight
