@@ -805,7 +805,7 @@ torch::Tensor stem_conv_wrw(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(Ci == 3 && (Co == 16 || Co == 32),
               "stem_wrw: supported shapes are Ci=3, Co in {16,32}");
   const int wn = Co * 9 * Ci;
-  const int B = 128;
+  const int B = 1024;
   auto f32 = x.options().dtype(torch::kFloat);
   auto part = torch::empty({B, wn}, f32);
   auto dw = torch::empty(
@@ -832,7 +832,8 @@ torch::Tensor stem_conv_wrw(torch::Tensor dy, torch::Tensor x) {
                            part.data_ptr<float>(), N, H, W);
     });
   });
-  hipLaunchKernelGGL(stem_wrw_final_k, dim3((wn + FT_BLOCK - 1) / FT_BLOCK),
+  hipLaunchKernelGGL(stem_wrw_final_k,
+                     dim3((wn + FT_BLOCK / WAVE - 1) / (FT_BLOCK / WAVE)),
                      dim3(FT_BLOCK), 0, STREAM, part.data_ptr<float>(), B,
                      wn, dw.data_ptr<float>());
   return dw;

@@ -26,7 +26,7 @@
 // one pixel.  Weights staged in LDS as [kh][kw][ci][co] so the co loop is
 // an LDS broadcast.
 template <typename TX, typename TY, int CI, int CO>
-__global__ void stem_fwd_k(const TX* __restrict__ x,
+__global__ void __launch_bounds__(FT_BLOCK) stem_fwd_k(const TX* __restrict__ x,
                            const float* __restrict__ w,
                            TY* __restrict__ y, int N, int H, int W) {
   __shared__ float wl[3][3][CI][CO];
@@ -37,14 +37,14 @@ __global__ void stem_fwd_k(const TX* __restrict__ x,
     wl[kh][kw][ci][co] = w[t];
   }
   __syncthreads();
-  const int HW = H * W;
-  const long total = (long)N * HW;
-  const long stride = (long)gridDim.x * blockDim.x;
-  for (long pix = blockIdx.x * (long)blockDim.x + threadIdx.x; pix < total;
+  const unsigned HW = H * W;
+  const unsigned total = (unsigned)N * HW;
+  const unsigned stride = gridDim.x * blockDim.x;
+  for (unsigned pix = blockIdx.x * blockDim.x + threadIdx.x; pix < total;
        pix += stride) {
-    const int hw = (int)(pix % HW);
-    const int n = (int)(pix / HW);
-    const int h = hw / W, ww_ = hw % W;
+    const unsigned hw = pix % HW;
+    const unsigned n = pix / HW;
+    const int h = hw / (unsigned)W, ww_ = hw % (unsigned)W;
     float acc[CO];
 #pragma unroll
     for (int co = 0; co < CO; ++co) acc[co] = 0.f;
@@ -67,7 +67,7 @@ __global__ void stem_fwd_k(const TX* __restrict__ x,
         }
       }
     }
-    TY* yp = y + pix * CO;
+    TY* yp = y + (long)pix * CO;
 #pragma unroll
     for (int co = 0; co < CO; ++co) yp[co] = (TY)acc[co];
   }
@@ -78,7 +78,7 @@ __global__ void stem_fwd_k(const TX* __restrict__ x,
 // reduction per block, then one partial row per block; stem_wrw_final_k
 // sums the partials.
 template <typename TY, typename TX, int CI, int CO>
-__global__ void stem_wrw_k(const TY* __restrict__ dy,
+__global__ void __launch_bounds__(FT_BLOCK) stem_wrw_k(const TY* __restrict__ dy,
                            const TX* __restrict__ x,
                            float* __restrict__ part /*[B, CO*9*CI]*/, int N,
                            int H, int W) {
@@ -86,18 +86,18 @@ __global__ void stem_wrw_k(const TY* __restrict__ dy,
   for (int t = threadIdx.x; t < CO * 9 * CI; t += blockDim.x) lds[t] = 0.f;
   __syncthreads();
   const int co = threadIdx.x % CO;  // CO divides 256
-  const int HW = H * W;
-  const long total = (long)N * HW;
-  const long pix0 = (blockIdx.x * (long)blockDim.x + threadIdx.x) / CO;
-  const long pstride = ((long)gridDim.x * blockDim.x) / CO;
+  const unsigned HW = H * W;
+  const unsigned total = (unsigned)N * HW;
+  const unsigned pix0 = (blockIdx.x * blockDim.x + threadIdx.x) / CO;
+  const unsigned pstride = (gridDim.x * blockDim.x) / CO;
   float acc[9 * CI];
 #pragma unroll
   for (int j = 0; j < 9 * CI; ++j) acc[j] = 0.f;
-  for (long pix = pix0; pix < total; pix += pstride) {
-    const int hw = (int)(pix % HW);
-    const int n = (int)(pix / HW);
-    const int h = hw / W, ww_ = hw % W;
-    const float g = (float)dy[pix * CO + co];
+  for (unsigned pix = pix0; pix < total; pix += pstride) {
+    const unsigned hw = pix % HW;
+    const unsigned n = pix / HW;
+    const int h = hw / (unsigned)W, ww_ = hw % (unsigned)W;
+    const float g = (float)dy[(long)pix * CO + co];
     const TX* xn = x + (long)n * HW * CI;
 #pragma unroll
     for (int kh = 0; kh < 3; ++kh) {
@@ -123,11 +123,17 @@ __global__ void stem_wrw_k(const TY* __restrict__ dy,
     part[(long)blockIdx.x * (CO * 9 * CI) + t] = lds[t];
 }
 
-__global__ void stem_wrw_final_k(const float* __restrict__ part, int B,
-                                 int wn, float* __restrict__ dw) {
-  const int t = blockIdx.x * blockDim.x + threadIdx.x;
+// one wave per output element: lane-parallel loads over the B partials +
+// xor-butterfly (a serial B-loop is a latency chain at B=1024).
+__global__ void __launch_bounds__(FT_BLOCK) stem_wrw_final_k(
+    const float* __restrict__ part, int B, int wn,
+    float* __restrict__ dw) {
+  const int wave = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
+  const int t = blockIdx.x * (FT_BLOCK / WAVE) + wave;
   if (t >= wn) return;
   float s = 0.f;
-  for (int b = 0; b < B; ++b) s += part[(long)b * wn + t];
-  dw[t] = s;
+  for (int b = lane; b < B; b += WAVE) s += part[(long)b * wn + t];
+#pragma unroll
+  for (int off = WAVE / 2; off; off >>= 1) s += __shfl_xor(s, off, WAVE);
+  if (lane == 0) dw[t] = s;
 }
