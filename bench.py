@@ -48,7 +48,7 @@ def parse():
     # NCHW measured faster than NHWC under hipGraphs on gfx950 (MIOpen falls
     # back to a naive NHWC wrw kernel for the 3-channel stem; see
     # profiles/r01_bench_notes.md)
-    p.add_argument('--layout', type=str, default='nchw',
+    p.add_argument('--layout', type=str, default='nhwc',
                    choices=['nhwc', 'nchw'])
     p.add_argument('--fused_bn', type=str, default='on',
                    choices=['on', 'off'])

@@ -27,6 +27,8 @@ def _load_data_batch(args, _input, _target):
     if args.graph.on_cuda and torch.cuda.is_available():
         _input = _input.cuda(non_blocking=True)
         _target = _target.cuda(non_blocking=True)
+        if getattr(args, 'channels_last', False) and _input.dim() == 4:
+            _input = _input.contiguous(memory_format=torch.channels_last)
     return _input, _target
 
 

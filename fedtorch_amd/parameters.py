@@ -204,7 +204,10 @@ def build_parser():
     parser.add_argument('--clients_per_rank', default=1, type=int,
                         help='virtual clients packed per GPU rank (their '
                              'replicas and aux state stay resident in HBM3E).')
-    parser.add_argument('--channels_last', type=str2bool, default=False)
+    # NHWC is the native layout on MI355X: MIOpen's igemm solvers are NHWC
+    # (NCHW tensors pay batched_transpose wrappers around every conv) and
+    # the fused BN / stem kernels have NHWC paths (profiles/r01_bench_notes.md)
+    parser.add_argument('--channels_last', type=str2bool, default=True)
     parser.add_argument('--aggregate_bn_stats', type=str2bool, default=True,
                         help='average BatchNorm running stats across online '
                              'clients at sync (the reference never syncs '
