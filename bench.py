@@ -45,11 +45,15 @@ def parse():
     p.add_argument('--dtype', type=str, default='bf16')
     p.add_argument('--graph', type=str, default='auto',
                    help='hipGraph-capture the local step: auto|on|off')
-    # NCHW measured faster than NHWC under hipGraphs on gfx950 (MIOpen falls
-    # back to a naive NHWC wrw kernel for the 3-channel stem; see
-    # profiles/r01_bench_notes.md)
+    # NHWC end-to-end (custom NHWC fused-BN + stem kernels dodge MIOpen's
+    # naive bf16 fallbacks) measured +23% over NCHW at b256; see
+    # profiles/r01_bench_notes.md
     p.add_argument('--layout', type=str, default='nhwc',
                    choices=['nhwc', 'nchw'])
+    p.add_argument('--bf16_weights', type=str, default='on',
+                   choices=['on', 'off'],
+                   help='bf16 compute twin for conv/linear weights (fp32 '
+                        'master in the arena; graph mode only)')
     p.add_argument('--fused_bn', type=str, default='on',
                    choices=['on', 'off'])
     p.add_argument('--fp32_stem', type=str, default='off',
@@ -155,13 +159,29 @@ def main():
     use_graph = on_gpu and b.graph != 'off'
     if use_graph:
         try:
+            arena = client.arena
+            if b.bf16_weights == 'on':
+                # conv/linear params live in a bf16 twin arena: no autocast
+                # weight-cast kernels per step; the fused SGD refreshes the
+                # twin and the grad gather casts bf16 grads back to fp32.
+                arena.enable_bf16_compute()
+
+            def inner_stolen(x, y):
+                arena.detach_grads()
+                with amp(args):
+                    loss = client.criterion(client.model(x), y)
+                loss.backward()
+                arena.gather_grads()
+                client.optimizer.step(apply_lr=True, apply_in_momentum=True,
+                                      apply_out_momentum=False)
+
             static_x = (xs[0] if isinstance(xs, list) else xs[0]).clone()
             static_y = ys[0].clone()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 for _ in range(3):
-                    inner(static_x, static_y)
+                    inner_stolen(static_x, static_y)
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             # Two captures with stolen grads: with p.grad=None at capture,
@@ -169,7 +189,6 @@ def main():
             # of launching one fp32 add per parameter per step (~65 kernels
             # on ResNet-20); graph2 gathers the stolen buffers into the
             # contiguous grad arena with ONE kernel, then steps.
-            arena = client.arena
             arena.detach_grads()
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
@@ -188,6 +207,7 @@ def main():
             print('[bench] hipGraph capture failed (%r), eager path' % e,
                   flush=True)
             use_graph = False
+            client.arena.disable_bf16_compute()
             client.arena.attach_grads()
     if use_graph:
         def local_step(i):
@@ -212,6 +232,7 @@ def main():
                                online, work=client.work)
         aggregate_bn_buffers(args, client.comm, client.arena, online,
                              work=client.work)
+        client.arena.sync_half()
 
     def run(n_steps):
         for s in range(1, n_steps + 1):

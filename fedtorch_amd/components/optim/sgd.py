@@ -104,7 +104,8 @@ class FusedSGD(object):
             ctrl_server=self._ctrl_server if local else None,
             ctrl_client=self._ctrl_client if local else None,
             delta=self._delta if local else None,
-            wd_numel=self.arena.wd_numel)
+            wd_numel=self.arena.wd_numel,
+            half_param=getattr(self.arena, 'half_flat', None))
         if use_in:
             self._in_init = True
         if use_out:

@@ -56,7 +56,8 @@ def fused_sgd_step(param, grad, *, lr, scale, weight_decay, in_momentum,
                    apply_in_momentum, apply_out_momentum,
                    in_buf=None, out_buf=None, first_in=False, first_out=False,
                    prox_mu=0.0, server=None, ctrl_server=None,
-                   ctrl_client=None, delta=None, wd_numel=None):
+                   ctrl_client=None, delta=None, wd_numel=None,
+                   half_param=None):
     """One pass over the arena: the reference's dual-use SGD step with the
     per-algorithm gradient corrections fused in.
 
@@ -90,7 +91,9 @@ def fused_sgd_step(param, grad, *, lr, scale, weight_decay, in_momentum,
             bool(nesterov), bool(apply_lr), bool(apply_in_momentum),
             bool(apply_out_momentum), bool(first_in), bool(first_out),
             float(prox_mu),
-            int(wd_numel) if wd_numel is not None else param.numel())
+            int(wd_numel) if wd_numel is not None else param.numel(),
+            half_param if half_param is not None
+            else torch.empty(0, device=param.device))
 
     d = grad
     if delta is not None:
@@ -123,6 +126,8 @@ def fused_sgd_step(param, grad, *, lr, scale, weight_decay, in_momentum,
         else:
             d = out_buf.clone()
     param.add_(d, alpha=-(lr if apply_lr else scale))
+    if half_param is not None:  # refresh the bf16 compute copy
+        half_param.copy_(param)
 
 
 # --------------------------------------------------------------------------

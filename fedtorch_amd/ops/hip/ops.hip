@@ -24,7 +24,7 @@
 enum {
   F_DELTA = 1, F_CTRL = 2, F_PROX = 4, F_WD = 8,
   F_IN = 16, F_OUT = 32, F_NESTEROV = 64, F_FIRST_IN = 128,
-  F_FIRST_OUT = 256
+  F_FIRST_OUT = 256, F_HALF = 512
 };
 
 __global__ void fused_sgd_kernel(
@@ -32,6 +32,7 @@ __global__ void fused_sgd_kernel(
     float* __restrict__ bin, float* __restrict__ bout,
     const float* __restrict__ delta, const float* __restrict__ cs,
     const float* __restrict__ cc, const float* __restrict__ server,
+    __hip_bfloat16* __restrict__ half_p,
     long n4, long wd_n4, float wd, float m_in, float m_out,
     float omd_in, float omd_out, float step_scale, float prox_mu,
     int flags) {
@@ -101,6 +102,13 @@ __global__ void fused_sgd_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) pp[j] = fmaf(-step_scale, d[j], pp[j]);
     reinterpret_cast<float4*>(p)[i] = pv;
+    if (flags & F_HALF) {  // refresh the bf16 compute copy in the same pass
+      struct alignas(8) H4 { __hip_bfloat16 d[4]; };
+      H4 hv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) hv.d[j] = __float2bfloat16(pp[j]);
+      reinterpret_cast<H4*>(half_p)[i] = hv;
+    }
   }
 }
 
@@ -111,7 +119,7 @@ void fused_sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor in_buf,
                     double m_in, double m_out, double damp, bool nesterov,
                     bool apply_lr, bool apply_in, bool apply_out,
                     bool first_in, bool first_out, double prox_mu,
-                    long wd_numel) {
+                    long wd_numel, torch::Tensor half_p) {
   CHK(p); CHK(g);
   TORCH_CHECK(p.numel() % 4 == 0, "arena numel must be float4-aligned");
   long n4 = p.numel() / 4;
@@ -125,6 +133,7 @@ void fused_sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor in_buf,
   if (nesterov) flags |= F_NESTEROV;
   if (first_in) flags |= F_FIRST_IN;
   if (first_out) flags |= F_FIRST_OUT;
+  if (half_p.numel()) flags |= F_HALF;
   hipLaunchKernelGGL(fused_sgd_kernel, dim3(ft_grid(n4)), dim3(FT_BLOCK), 0,
                      STREAM, p.data_ptr<float>(), g.data_ptr<float>(),
                      (flags & F_IN) ? in_buf.data_ptr<float>() : nullptr,
@@ -133,6 +142,9 @@ void fused_sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor in_buf,
                      (flags & F_CTRL) ? ctrl_server.data_ptr<float>() : nullptr,
                      (flags & F_CTRL) ? ctrl_client.data_ptr<float>() : nullptr,
                      (flags & F_PROX) ? server.data_ptr<float>() : nullptr,
+                     (flags & F_HALF) ? reinterpret_cast<__hip_bfloat16*>(
+                                            half_p.data_ptr())
+                                      : nullptr,
                      n4, wd_numel / 4, (float)wd, (float)m_in, (float)m_out,
                      (float)(1.0 - damp), (float)(1.0 - damp),
                      (float)(apply_lr ? lr : scale), (float)prox_mu, flags);
@@ -708,15 +720,39 @@ void scatter_accumulate(torch::Tensor out, torch::Tensor vs,
 // the grad-norm trackers read.  Chunk table is built once per capture on
 // the host (shapes are static).
 // ==========================================================================
+// table row: (src_idx, src_off, dst_off, n, is_bf16).  bf16 sources (grads
+// of bf16-compute params) are cast to fp32 during the copy — no separate
+// cast kernel.
 __global__ void gather_chunks_kernel(const unsigned long long* __restrict__
                                          srcs,
-                                     const int* __restrict__ table /*[B,4]*/,
+                                     const int* __restrict__ table /*[B,5]*/,
                                      float* __restrict__ dst) {
-  const int* e = table + 4 * blockIdx.x;
-  const float* src =
-      reinterpret_cast<const float*>(srcs[e[0]]) + e[1];
+  const int* e = table + 5 * blockIdx.x;
   float* out = dst + e[2];
   const int n = e[3];
+  if (e[4]) {  // bf16 source
+    const __hip_bfloat16* src =
+        reinterpret_cast<const __hip_bfloat16*>(srcs[e[0]]) + e[1];
+    const int n8 = n >> 3;
+    for (int i = threadIdx.x; i < n8; i += blockDim.x) {
+      struct alignas(16) V8 { __hip_bfloat16 d[8]; };
+      V8 v = reinterpret_cast<const V8*>(src)[i];
+      float4 lo, hi;
+      float* l = reinterpret_cast<float*>(&lo);
+      float* h = reinterpret_cast<float*>(&hi);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        l[j] = __bfloat162float(v.d[j]);
+        h[j] = __bfloat162float(v.d[4 + j]);
+      }
+      reinterpret_cast<float4*>(out)[2 * i] = lo;
+      reinterpret_cast<float4*>(out)[2 * i + 1] = hi;
+    }
+    for (int i = (n8 << 3) + threadIdx.x; i < n; i += blockDim.x)
+      out[i] = __bfloat162float(src[i]);
+    return;
+  }
+  const float* src = reinterpret_cast<const float*>(srcs[e[0]]) + e[1];
   const int n4 = n >> 2;
   for (int i = threadIdx.x; i < n4; i += blockDim.x)
     reinterpret_cast<float4*>(out)[i] =
@@ -728,7 +764,7 @@ __global__ void gather_chunks_kernel(const unsigned long long* __restrict__
 void gather_grads(torch::Tensor srcs, torch::Tensor table,
                   torch::Tensor dst) {
   CHK(dst); CHK(srcs); CHK(table);
-  TORCH_CHECK(table.dim() == 2 && table.size(1) == 4, "table must be [B,4]");
+  TORCH_CHECK(table.dim() == 2 && table.size(1) == 5, "table must be [B,5]");
   TORCH_CHECK(srcs.scalar_type() == torch::kLong, "srcs must be int64 ptrs");
   TORCH_CHECK(table.scalar_type() == torch::kInt, "table must be int32");
   const int B = (int)table.size(0);
@@ -737,6 +773,38 @@ void gather_grads(torch::Tensor srcs, torch::Tensor table,
                      reinterpret_cast<const unsigned long long*>(
                          srcs.data_ptr<long>()),
                      table.data_ptr<int>(), dst.data_ptr<float>());
+}
+
+// fp32 master arena -> bf16 compute arena (one pass; used after any direct
+// mutation of the master outside the fused SGD kernel, e.g. aggregation)
+__global__ void cast_half_kernel(const float* __restrict__ src,
+                                 __hip_bfloat16* __restrict__ dst, long n8) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    float4 lo = reinterpret_cast<const float4*>(src)[2 * i];
+    float4 hi = reinterpret_cast<const float4*>(src)[2 * i + 1];
+    struct alignas(16) V8 { __hip_bfloat16 d[8]; };
+    V8 v;
+    const float* l = reinterpret_cast<const float*>(&lo);
+    const float* h = reinterpret_cast<const float*>(&hi);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v.d[j] = __float2bfloat16(l[j]);
+      v.d[4 + j] = __float2bfloat16(h[j]);
+    }
+    reinterpret_cast<V8*>(dst)[i] = v;
+  }
+}
+
+void cast_to_half(torch::Tensor src, torch::Tensor dst) {
+  CHK(src); CHK(dst);
+  TORCH_CHECK(src.numel() == dst.numel() && src.numel() % 8 == 0,
+              "cast_to_half: numel mismatch / not 8-aligned");
+  long n8 = src.numel() / 8;
+  hipLaunchKernelGGL(cast_half_kernel, dim3(ft_grid(n8)), dim3(FT_BLOCK), 0,
+                     STREAM, src.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dst.data_ptr()), n8);
 }
 
 // ==========================================================================
@@ -1083,4 +1151,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_grads", &gather_grads);
   m.def("stem_conv_fwd", &stem_conv_fwd);
   m.def("stem_conv_wrw", &stem_conv_wrw);
+  m.def("cast_to_half", &cast_to_half);
 }

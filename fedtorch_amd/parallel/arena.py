@@ -122,6 +122,69 @@ class Arena(object):
         N, C, H, W = shape
         return flat[off:off + n].view(N, H, W, C).permute(0, 3, 1, 2)
 
+    # ---- bf16 compute copy (fp32 master) -----------------------------------
+    # Conv/Linear weights are what autocast casts to bf16 EVERY step (~48
+    # cast kernels per ResNet-20 step under hipGraph replay).  Instead, those
+    # parameters point at a bf16 twin of the arena; the fused SGD kernel
+    # refreshes the twin in its own pass over the fp32 master, and the grad
+    # gather casts the (now bf16) stolen grads back to fp32 during its copy.
+    # Intended for the hipGraph/stolen-grad path (autograd requires p.grad
+    # dtype == p dtype, so attached-view mode keeps fp32 params).
+    half_flat = None
+
+    def enable_bf16_compute(self):
+        """Re-point conv (4D) and nn.Linear parameters at a bf16 twin
+        arena.  Returns self."""
+        if self.half_flat is not None:
+            return self
+        import torch.nn as nn
+        linear_params = set()
+        for m in self.module.modules():
+            if isinstance(m, nn.Linear):
+                for p in m.parameters(recurse=False):
+                    linear_params.add(id(p))
+        self.half_flat = torch.zeros(self.numel, dtype=torch.bfloat16,
+                                     device=self.device)
+        self.half_flat.copy_(self.flat)
+        self.bf16 = []
+        for p, off, n, cl in zip(self.params, self.offsets, self.numels,
+                                 self.cl):
+            use = p.dim() == 4 or id(p) in linear_params
+            self.bf16.append(use)
+            if not use:
+                continue
+            if cl:
+                p.data = self._cl_view(self.half_flat, off, n, p.shape)
+            else:
+                p.data = self.half_flat[off:off + n].view(p.shape)
+        self._gather_state = None
+        return self
+
+    def disable_bf16_compute(self):
+        """Restore fp32 master views (e.g. when falling back to eager)."""
+        if self.half_flat is None:
+            return
+        for p, off, n, cl in zip(self.params, self.offsets, self.numels,
+                                 self.cl):
+            if cl:
+                p.data = self._cl_view(self.flat, off, n, p.shape)
+            else:
+                p.data = self.flat[off:off + n].view(p.shape)
+        self.half_flat = None
+        self.bf16 = None
+        self._gather_state = None
+
+    def sync_half(self):
+        """Refresh the bf16 twin after a direct mutation of the fp32 master
+        (aggregation writes outside the fused SGD kernel)."""
+        if self.half_flat is None:
+            return
+        from fedtorch_amd import ops
+        if ops.hip_available() and self.flat.is_cuda:
+            ops._C.cast_to_half(self.flat, self.half_flat)
+        else:
+            self.half_flat.copy_(self.flat)
+
     # ---- gradient plumbing -------------------------------------------------
     def attach_grads(self):
         """Point every p.grad at its view of the grad arena."""
@@ -160,11 +223,13 @@ class Arena(object):
                      and (g.is_contiguous(
                              memory_format=torch.channels_last)
                           if cl else g.is_contiguous()))
-            if not dense or g.dtype != torch.float32:
+            if not dense or g.dtype not in (torch.float32, torch.bfloat16):
                 return None  # fall back to per-tensor copies
+            bf16 = 1 if g.dtype == torch.bfloat16 else 0
             srcs.append(g.data_ptr())
             for c in range(0, n, self._GATHER_CHUNK):
-                table.append((i, c, off + c, min(self._GATHER_CHUNK, n - c)))
+                table.append((i, c, off + c, min(self._GATHER_CHUNK, n - c),
+                              bf16))
         dev = self.flat.device
         return (torch.tensor(srcs, dtype=torch.int64, device=dev),
                 torch.tensor(table, dtype=torch.int32, device=dev))
@@ -180,11 +245,11 @@ class Arena(object):
             from fedtorch_amd import ops
             ops._C.gather_grads(self._gather_state[0],
                                 self._gather_state[1], self.grad)
-        else:  # non-contiguous / non-fp32 stolen grads: plain copies
+        else:  # non-contiguous / exotic-dtype stolen grads: plain copies
             for p, off, n, cl in zip(self.params, self.offsets, self.numels,
                                      self.cl):
                 g = p.grad.permute(0, 2, 3, 1) if cl else p.grad
-                self.grad[off:off + n].copy_(g.reshape(-1))
+                self.grad[off:off + n].copy_(g.reshape(-1).float())
 
     # ---- flat state helpers ------------------------------------------------
     def new_buffer(self, zero=True):
@@ -205,11 +270,19 @@ class Arena(object):
                 for off, n, shape in zip(self.offsets, self.numels, self.shapes)]
 
     def check_views(self):
-        """True if every param still points into the arena (a module-level
-        ``p.data = ...`` assignment elsewhere would silently detach it)."""
-        ptr0 = self.flat.data_ptr()
-        end = ptr0 + self.flat.numel() * self.flat.element_size()
-        return all(ptr0 <= p.data_ptr() < end for p in self.params)
+        """True if every param still points into the arena — or, in bf16
+        compute mode, into the bf16 twin (a module-level ``p.data = ...``
+        assignment elsewhere would silently detach it)."""
+        ranges = [(self.flat.data_ptr(),
+                   self.flat.data_ptr()
+                   + self.flat.numel() * self.flat.element_size())]
+        if self.half_flat is not None:
+            ranges.append((self.half_flat.data_ptr(),
+                           self.half_flat.data_ptr()
+                           + self.half_flat.numel()
+                           * self.half_flat.element_size()))
+        return all(any(lo <= p.data_ptr() < hi for lo, hi in ranges)
+                   for p in self.params)
 
     # ---- interop -----------------------------------------------------------
     def state_dict_flat(self):

@@ -478,6 +478,53 @@ def test_nhwc_resnet20_step_matches_nchw():
     assert abs(n_cl - n_ref) / max(n_ref, 1e-6) < 0.1, (n_ref, n_cl)
 
 
+def test_bf16_compute_arena_step_matches_fp32():
+    """bf16 compute twin (fp32 master): a fwd/bwd/step through bf16 views +
+    gather tracks the fp32+autocast path, and the twin stays in sync."""
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    from fedtorch_amd.components.optim.sgd import FusedSGD
+
+    def build():
+        torch.manual_seed(9)
+        return nn.Sequential(nn.Linear(32, 64), nn.ReLU(),
+                             nn.Linear(64, 10)).cuda()
+
+    x = torch.randn(64, 32, device='cuda')
+    y = torch.randint(0, 10, (64,), device='cuda')
+    crit = nn.CrossEntropyLoss()
+
+    m1 = build()
+    a1 = Arena(m1)
+    o1 = FusedSGD(a1, lr=0.1)
+    a1.zero_grad()
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        crit(m1(x), y).backward()
+    o1.step(apply_lr=True)
+
+    m2 = build()
+    a2 = Arena(m2)
+    o2 = FusedSGD(a2, lr=0.1)
+    a2.enable_bf16_compute()
+    assert m2[0].weight.dtype == torch.bfloat16
+    a2.detach_grads()
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        crit(m2(x), y).backward()
+    a2.gather_grads()
+    o2.step(apply_lr=True)
+    torch.cuda.synchronize()
+    # masters agree to bf16-grad noise
+    assert torch.allclose(a2.flat, a1.flat, atol=2e-3), \
+        (a2.flat - a1.flat).abs().max().item()
+    # twin refreshed by the fused step in the same pass
+    assert torch.allclose(a2.half_flat.float(), a2.flat, atol=1e-2)
+    # direct master mutation + sync_half
+    a2.flat.mul_(0.5)
+    a2.sync_half()
+    torch.cuda.synchronize()
+    assert torch.allclose(a2.half_flat.float(), a2.flat, atol=1e-2)
+
+
 def test_fused_bn_nbt_lazy_flush():
     """num_batches_tracked counts on the host and flushes into the buffer
     when the state_dict is read (no per-step GPU kernel)."""
