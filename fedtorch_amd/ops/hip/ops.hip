@@ -943,14 +943,23 @@ static inline int bnh_lgc(long C, int VN) {
   for (long c = C / VN; c > 1; c >>= 1) ++l;
   return l;
 }
+// grid caps overridable via env for on-box tuning sweeps (read once)
+static inline int ft_env_int(const char* k, int d) {
+  const char* v = getenv(k);
+  return v ? atoi(v) : d;
+}
 static inline int bnh_red_grid(long tasks) {
-  long b = (tasks + FT_BLOCK * 8 - 1) / (FT_BLOCK * 8);
-  if (b > 128) b = 128;
+  static const int cap = ft_env_int("FT_BNH_RED_CAP", 128);
+  static const int iters = ft_env_int("FT_BNH_RED_ITERS", 8);
+  long b = (tasks + FT_BLOCK * iters - 1) / (FT_BLOCK * iters);
+  if (b > cap) b = cap;
   return (int)(b < 1 ? 1 : b);
 }
 static inline int bnh_ew_grid(long tasks) {
-  long b = (tasks + FT_BLOCK * 2 - 1) / (FT_BLOCK * 2);
-  if (b > 1024) b = 1024;
+  static const int cap = ft_env_int("FT_BNH_EW_CAP", 1024);
+  static const int iters = ft_env_int("FT_BNH_EW_ITERS", 2);
+  long b = (tasks + FT_BLOCK * iters - 1) / (FT_BLOCK * iters);
+  if (b > cap) b = cap;
   return (int)(b < 1 ? 1 : b);
 }
 
