@@ -6,6 +6,8 @@ import os
 import subprocess
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 SHAPES = [(256, 16, 32, 32), (256, 32, 16, 16), (256, 64, 8, 8)]
 
 
