@@ -209,7 +209,11 @@ class FederatedPartitioner(Partitioner):
         num_classes = len(class_array)
         probs = np.random.dirichlet(num_classes * [0.1 / num_classes], n)
         probs[probs * client_data_size < 10] = 0
-        probs = probs * class_sample_size / np.sum(probs, 0)
+        col = np.sum(probs, 0)
+        # a class no client drew (column sum 0) gets no samples instead of
+        # a 0/0 NaN -> invalid int cast
+        probs = np.divide(probs * class_sample_size, col,
+                          out=np.zeros_like(probs), where=col > 0)
         sample_sizes = probs.astype(int)
         ptr = np.zeros(num_classes, dtype=int)
         for client in range(n):

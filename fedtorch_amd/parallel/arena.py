@@ -238,9 +238,11 @@ class Arena(object):
         """One-kernel copy of the stolen per-parameter grads into the grad
         arena.  Builds the chunk table on first use (static shapes)."""
         if getattr(self, '_gather_state', None) is None:
-            self._gather_state = self._build_gather_state()
-            if self._gather_state is None:
-                self._gather_state = ()
+            from fedtorch_amd import ops
+            state = None
+            if self.flat.is_cuda and ops.hip_available():
+                state = self._build_gather_state()
+            self._gather_state = state if state is not None else ()
         if self._gather_state:
             from fedtorch_amd import ops
             ops._C.gather_grads(self._gather_state[0],

@@ -183,3 +183,55 @@ def test_bnrelu_deepcopy_safe():
     assert torch.allclose(m(x), c(x))
     # relu applied exactly once on the fallback path
     assert (c(x) >= 0).all()
+
+
+def test_arena_channels_last_packing():
+    """4D params of a channels_last module keep channels_last strides
+    through the arena, values round-trip, and the model still runs."""
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    torch.manual_seed(3)
+    m = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1, bias=False),
+                      nn.Conv2d(8, 8, 1, bias=False))
+    m = m.to(memory_format=torch.channels_last)
+    before = [p.detach().clone() for p in m.parameters()]
+    a = Arena(m)
+    cl = torch.channels_last
+    w0 = m[0].weight
+    assert w0.is_contiguous(memory_format=cl) and not w0.is_contiguous()
+    for p, b in zip(m.parameters(), before):
+        assert torch.equal(p.detach(), b)
+    assert a.check_views()
+    x = torch.randn(2, 3, 8, 8).contiguous(memory_format=cl)
+    y = m(x)
+    y.sum().backward()
+    # grads accumulated into the arena (attached views)
+    assert a.grad.abs().sum() > 0
+    # gather fallback path (CPU): detach + eager backward + gather copies
+    a.detach_grads()
+    a.zero_grad()
+    m(x).sum().backward()
+    a.gather_grads()
+    assert a.grad.abs().sum() > 0
+
+
+def test_arena_bf16_compute_cpu_roundtrip():
+    """enable/disable of the bf16 compute twin re-points params correctly
+    (CPU: values only; the fused-kernel path is covered by GPU tests)."""
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    torch.manual_seed(4)
+    m = nn.Sequential(nn.Linear(8, 8), nn.BatchNorm1d(8))
+    a = Arena(m)
+    ref = a.flat.clone()
+    a.enable_bf16_compute()
+    assert m[0].weight.dtype == torch.bfloat16
+    assert m[1].weight.dtype == torch.float32  # BN params stay fp32
+    assert a.check_views()
+    a.flat.mul_(2.0)
+    a.sync_half()
+    assert torch.allclose(a.half_flat.float(), a.flat, atol=1e-1)
+    a.disable_bf16_compute()
+    assert m[0].weight.dtype == torch.float32
+    assert torch.allclose(a.flat, ref * 2.0)
+    assert a.check_views()
