@@ -8,8 +8,9 @@ import sys
 
 
 def summarize(db_path, out_path, top=30, window=0.0):
-    """window: fraction of the timeline to SKIP from the start (e.g. 0.5
-    keeps only the second half — steady state, past MIOpen find/warmup)."""
+    """window <= 1: fraction of the timeline to SKIP from the start.
+    window > 1: keep only the LAST `window` MILLISECONDS (robust steady-state
+    selection when MIOpen-find dominates the wall)."""
     db = sqlite3.connect(db_path)
     cur = db.cursor()
     tables = [r[0] for r in cur.execute(
@@ -21,13 +22,17 @@ def summarize(db_path, out_path, top=30, window=0.0):
     t0, t1 = cur.execute(
         "SELECT MIN(start), MAX(end) FROM rocpd_kernel_dispatch_%s"
         % sfx).fetchone()
-    cut = t0 + (t1 - t0) * window
+    if window > 1.0:
+        cut = t1 - window * 1e6  # rocprof timestamps are ns
+    else:
+        cut = t0 + (t1 - t0) * window
     where = "WHERE CAST(k.start AS REAL) >= %d" % cut
     tot = cur.execute(
         "SELECT SUM(k.end-k.start)/1e6, COUNT(*), "
         "(MAX(k.end)-MIN(k.start))/1e6 "
         "FROM rocpd_kernel_dispatch_%s k %s" % (sfx, where)).fetchone()
-    lines = ['db: %s (window skip=%.0f%%)' % (db_path, window * 100),
+    hdr = ('last %.0f ms' % window) if window > 1 else ('skip=%.0f%%' % (window * 100))
+    lines = ['db: %s (window %s)' % (db_path, hdr),
              'TOTAL kernel %.1f ms / %d dispatches; wall span %.1f ms' % tot]
     q = ("SELECT ks.display_name, COUNT(*), SUM(k.end-k.start)/1e6, "
          "AVG(k.end-k.start)/1e3 FROM rocpd_kernel_dispatch_%s k "
