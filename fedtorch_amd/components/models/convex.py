@@ -65,6 +65,36 @@ class LeastSquare(nn.Module):
         return self.fc(x)
 
 
+class LinearMAFL(nn.Module):
+    """Two-factor linear model W@Z (reference
+    `models/convex/least_square.py:43-67`): a server core `W` over a
+    client-side projection `Z`, with the effective dense weight exposed as
+    ``.weight``.  NOTE: in the reference this is dead code — no
+    `--federated_type mafl` flag or training loop exists (`parameters.py`
+    never defines `mafl_server_dim`); kept for API parity."""
+
+    def __init__(self, in_features, middle_features, out_features=1):
+        super().__init__()
+        self.in_features = in_features if in_features else middle_features
+        self.middle_features = middle_features
+        self.out_features = out_features
+        self.Z = nn.Linear(self.in_features, middle_features, bias=False)
+        self.W = nn.Linear(middle_features, out_features, bias=True)
+        self.core_params = self.W.parameters()
+        self.extra_params = self.Z.parameters()
+
+    @property
+    def weight(self):
+        return torch.matmul(self.W.weight, self.Z.weight)
+
+    @property
+    def bias(self):
+        return self.W.bias
+
+    def forward(self, x):
+        return self.W(self.Z(x))
+
+
 def logistic_regression(args):
     return LogisticRegression(dataset=args.data)
 
@@ -74,6 +104,10 @@ def robust_logistic_regression(args):
 
 
 def least_square(args):
+    if getattr(args, 'federated_type', None) == 'mafl':
+        return LinearMAFL(in_features=getattr(args, 'input_dim', 0) or 0,
+                          middle_features=getattr(args, 'mafl_server_dim',
+                                                  10))
     return LeastSquare(dataset=args.data)
 
 
