@@ -338,20 +338,42 @@ def run_packed_bench(b, client, args, xs, ys, pool_n, world, rank, on_gpu):
         sl['opt'].step(apply_lr=True, apply_in_momentum=True,
                        apply_out_momentum=False)
 
+    def slot_step_stolen(sl):
+        # stolen-grad flow (same scheme as the single-client graph path)
+        sl['arena'].detach_grads()
+        with amp(args):
+            loss = client.criterion(sl['model'](sl['sx']), sl['sy'])
+        loss.backward()
+        sl['arena'].gather_grads()
+        sl['opt'].step(apply_lr=True, apply_in_momentum=True,
+                       apply_out_momentum=False)
+
     if on_gpu and b.graph != 'off':
+        if b.bf16_weights == 'on':
+            for sl in slots:
+                sl['arena'].enable_bf16_compute()
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for sl in slots:
                 for _ in range(3):
-                    slot_step(sl)
+                    slot_step_stolen(sl)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         for sl in slots:
-            gph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(gph):
-                slot_step(sl)
-            sl['graph'] = gph
+            sl['arena'].detach_grads()
+            g1 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g1):
+                with amp(args):
+                    loss = client.criterion(sl['model'](sl['sx']), sl['sy'])
+                loss.backward()
+            sl['arena'].gather_grads()
+            g2 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g2):
+                sl['arena'].gather_grads()
+                sl['opt'].step(apply_lr=True, apply_in_momentum=True,
+                               apply_out_momentum=False)
+            sl['graph'] = (g1, g2)
 
     main_stream = torch.cuda.current_stream() if on_gpu else None
 
@@ -373,12 +395,14 @@ def run_packed_bench(b, client, args, xs, ys, pool_n, world, rank, on_gpu):
                     sl['stream'].wait_stream(main_stream)
                     with torch.cuda.stream(sl['stream']):
                         sl['arena'].flat.copy_(server)
+                        sl['arena'].sync_half()
                         sl['opt']._in_buf.copy_(mom[c])
                         for t in range(TAU):
                             sl['sx'].copy_(xs[(c + t) % pool_n])
                             sl['sy'].copy_(ys[(c + t) % pool_n])
                             if sl['graph'] is not None:
-                                sl['graph'].replay()
+                                sl['graph'][0].replay()
+                                sl['graph'][1].replay()
                             else:
                                 slot_step(sl)
                         state[c].copy_(sl['arena'].flat)
