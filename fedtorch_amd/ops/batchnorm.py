@@ -42,7 +42,7 @@ class _FusedBNFunction(torch.autograd.Function):
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
                 eps, relu, res):
         empty = torch.empty(0, device=x.device)
-        y, save_mean, save_ivar = ops._C.bn_fwd_train(
+        y, save_mean, save_ivar, mask = ops._C.bn_fwd_train(
             x, weight if weight is not None else empty,
             bias if bias is not None else empty,
             running_mean if running_mean is not None else empty,
@@ -52,7 +52,11 @@ class _FusedBNFunction(torch.autograd.Function):
         ctx.relu = bool(relu)
         ctx.has_res = res is not None
         ctx.nhwc = _is_cl(x)
-        if relu:
+        ctx.has_mask = mask.numel() > 0
+        if ctx.has_mask:
+            # ReLU bitmask replaces the saved output (bwd reads 1 bit/elem)
+            ctx.save_for_backward(x, weight, save_mean, save_ivar, mask)
+        elif relu:
             ctx.save_for_backward(x, weight, save_mean, save_ivar, y)
         else:
             ctx.save_for_backward(x, weight, save_mean, save_ivar)
@@ -60,7 +64,11 @@ class _FusedBNFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        if ctx.relu:
+        mask = None
+        if ctx.has_mask:
+            x, weight, save_mean, save_ivar, mask = ctx.saved_tensors
+            y = torch.empty(0, device=x.device)
+        elif ctx.relu:
             x, weight, save_mean, save_ivar, y = ctx.saved_tensors
         else:
             x, weight, save_mean, save_ivar = ctx.saved_tensors
@@ -71,7 +79,9 @@ class _FusedBNFunction(torch.autograd.Function):
             dy = dy.contiguous()
         empty = torch.empty(0, device=x.device)
         dx, dweight, dbias, dres = ops._C.bn_bwd(
-            dy, x, y, save_mean, save_ivar,
+            dy, x, y, mask if mask is not None else
+            torch.empty(0, device=x.device, dtype=torch.uint8),
+            save_mean, save_ivar,
             weight if weight is not None else empty, ctx.relu, ctx.has_res)
         return (dx, dweight if weight is not None else None,
                 dbias if weight is not None else None, None, None, None,

@@ -293,7 +293,8 @@ __global__ void bnh_norm_k(const T* __restrict__ x, T* __restrict__ y,
                            float* __restrict__ save_ivar,
                            float* __restrict__ running_mean,
                            float* __restrict__ running_var,
-                           const T* __restrict__ res, long NI, long C,
+                           const T* __restrict__ res,
+                           unsigned char* __restrict__ mask, long NI, long C,
                            int lgc, float eps, float momentum, int relu) {
   // block-parallel finalize: thread (p, c) sums a slice of the B partials
   // so the L2 latencies overlap (a per-channel serial loop here is a
@@ -352,21 +353,27 @@ __global__ void bnh_norm_k(const T* __restrict__ x, T* __restrict__ y,
   for (unsigned r = t >> lgc; r < NI; r += rstride) {
     const long base = (long)r * C + cg * VN;
     VT v = *reinterpret_cast<const VT*>(x + base);
+    unsigned mbits = 0;
     if (res) {
       VT rv = *reinterpret_cast<const VT*>(res + base);
 #pragma unroll
       for (int j = 0; j < VN; ++j) {
         float f = fmaf((float)v.d[j], sc[j], sh[j]) + (float)rv.d[j];
+        if (f > 0.f) mbits |= 1u << j;
         v.d[j] = (T)(relu ? fmaxf(f, 0.f) : f);
       }
     } else {
 #pragma unroll
       for (int j = 0; j < VN; ++j) {
         float f = fmaf((float)v.d[j], sc[j], sh[j]);
+        if (f > 0.f) mbits |= 1u << j;
         v.d[j] = (T)(relu ? fmaxf(f, 0.f) : f);
       }
     }
     *reinterpret_cast<VT*>(y + base) = v;
+    // ReLU bitmask (VN==8): the bwd kernels read 1 byte per vector instead
+    // of re-reading the 16 B output vector for the mask
+    if (mask) mask[((long)r * C >> 3) + cg] = (unsigned char)mbits;
   }
 }
 
@@ -374,6 +381,7 @@ template <typename T, typename VT, int VN>
 __global__ void bnh_bwd_stats_k(const T* __restrict__ dy,
                                 const T* __restrict__ x,
                                 const T* __restrict__ yv,
+                                const unsigned char* __restrict__ mask,
                                 const float* __restrict__ save_mean,
                                 const float* __restrict__ save_ivar, long NI,
                                 long C, int lgc,
@@ -397,10 +405,17 @@ __global__ void bnh_bwd_stats_k(const T* __restrict__ dy,
     VT g = *reinterpret_cast<const VT*>(dy + base);
     VT xv = *reinterpret_cast<const VT*>(x + base);
     if (relu) {
-      VT yy = *reinterpret_cast<const VT*>(yv + base);
+      if (mask) {
+        const unsigned m = mask[((long)r * C >> 3) + cg];
 #pragma unroll
-      for (int j = 0; j < VN; ++j)
-        if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+        for (int j = 0; j < VN; ++j)
+          if (!((m >> j) & 1u)) g.d[j] = (T)0.f;
+      } else {
+        VT yy = *reinterpret_cast<const VT*>(yv + base);
+#pragma unroll
+        for (int j = 0; j < VN; ++j)
+          if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+      }
     }
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
@@ -446,6 +461,7 @@ template <typename T, typename VT, int VN>
 __global__ void bnh_bwd_dx_k(const T* __restrict__ dy,
                              const T* __restrict__ x,
                              const T* __restrict__ yv,
+                             const unsigned char* __restrict__ mask,
                              const float* __restrict__ part, int B,
                              const float* __restrict__ save_mean,
                              const float* __restrict__ save_ivar,
@@ -510,10 +526,17 @@ __global__ void bnh_bwd_dx_k(const T* __restrict__ dy,
     VT g = *reinterpret_cast<const VT*>(dy + base);
     VT xv = *reinterpret_cast<const VT*>(x + base);
     if (relu) {
-      VT yy = *reinterpret_cast<const VT*>(yv + base);
+      if (mask) {
+        const unsigned m = mask[((long)r * C >> 3) + cg];
 #pragma unroll
-      for (int j = 0; j < VN; ++j)
-        if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+        for (int j = 0; j < VN; ++j)
+          if (!((m >> j) & 1u)) g.d[j] = (T)0.f;
+      } else {
+        VT yy = *reinterpret_cast<const VT*>(yv + base);
+#pragma unroll
+        for (int j = 0; j < VN; ++j)
+          if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+      }
     }
     if (dres) *reinterpret_cast<VT*>(dres + base) = g;
 #pragma unroll

@@ -979,6 +979,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
   auto save_mean = torch::empty({C}, f32);
   auto save_ivar = torch::empty({C}, f32);
   auto y = torch::empty_like(x);
+  auto mask = torch::empty({0}, x.options().dtype(torch::kByte));
   bool track = running_mean.numel() > 0;
   if (nhwc) {
     AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
@@ -991,6 +992,9 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
       const long NI = N * HW, tasks = NI << lgc;
       const int B = bnh_red_grid(tasks);
       auto part = torch::empty({B, C, 2}, f32);
+      if (relu && VN == 8)  // ReLU bitmask: bwd reads 1 bit/elem, not y
+        mask = torch::empty({(NI * C) >> 3},
+                            x.options().dtype(torch::kByte));
       hipLaunchKernelGGL((bnh_stats_k<T, typename BnVec<T>::V, VN>),
                          dim3(B), dim3(FT_BLOCK), 0, STREAM,
                          reinterpret_cast<const T*>(x.data_ptr()), NI, C,
@@ -1009,10 +1013,12 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                          res.numel()
                              ? reinterpret_cast<const T*>(res.data_ptr())
                              : nullptr,
+                         mask.numel() ? mask.data_ptr<unsigned char>()
+                                      : nullptr,
                          NI, C, lgc, (float)eps, (float)momentum,
                          relu ? 1 : 0);
     });
-    return {y, save_mean, save_ivar};
+    return {y, save_mean, save_ivar, mask};
   }
   AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
                                  "bn_fwd", [&] {
@@ -1045,11 +1051,12 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                    : nullptr,
                        N, C, HW, (float)eps, (float)momentum, relu ? 1 : 0);
   });
-  return {y, save_mean, save_ivar};
+  return {y, save_mean, save_ivar, mask};
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
-                                  torch::Tensor y, torch::Tensor save_mean,
+                                  torch::Tensor y, torch::Tensor mask,
+                                  torch::Tensor save_mean,
                                   torch::Tensor save_ivar,
                                   torch::Tensor weight, bool relu,
                                   bool has_res) {
@@ -1077,18 +1084,23 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
       const long NI = N * HW, tasks = NI << lgc;
       const int B = bnh_red_grid(tasks);
       auto part = torch::empty({B, C, 2}, f32);
-      const T* yp = relu ? reinterpret_cast<const T*>(y.data_ptr()) : nullptr;
+      const unsigned char* mp =
+          mask.numel() ? mask.data_ptr<unsigned char>() : nullptr;
+      const T* yp = (relu && !mp)
+                        ? reinterpret_cast<const T*>(y.data_ptr())
+                        : nullptr;
+      TORCH_CHECK(!relu || mp || y.numel(), "bn_bwd NHWC: relu needs y/mask");
       hipLaunchKernelGGL((bnh_bwd_stats_k<T, typename BnVec<T>::V, VN>),
                          dim3(B), dim3(FT_BLOCK), 0, STREAM,
                          reinterpret_cast<const T*>(dy.data_ptr()),
-                         reinterpret_cast<const T*>(x.data_ptr()), yp,
+                         reinterpret_cast<const T*>(x.data_ptr()), yp, mp,
                          save_mean.data_ptr<float>(),
                          save_ivar.data_ptr<float>(), NI, C, lgc,
                          part.data_ptr<float>(), relu ? 1 : 0);
       hipLaunchKernelGGL((bnh_bwd_dx_k<T, typename BnVec<T>::V, VN>),
                          dim3(bnh_ew_grid(tasks)), dim3(FT_BLOCK), 0, STREAM,
                          reinterpret_cast<const T*>(dy.data_ptr()),
-                         reinterpret_cast<const T*>(x.data_ptr()), yp,
+                         reinterpret_cast<const T*>(x.data_ptr()), yp, mp,
                          part.data_ptr<float>(), B,
                          save_mean.data_ptr<float>(),
                          save_ivar.data_ptr<float>(),
