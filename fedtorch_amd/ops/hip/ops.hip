@@ -949,7 +949,7 @@ static inline int ft_env_int(const char* k, int d) {
   return v ? atoi(v) : d;
 }
 static inline int bnh_red_grid(long tasks) {
-  static const int cap = ft_env_int("FT_BNH_RED_CAP", 128);
+  static const int cap = ft_env_int("FT_BNH_RED_CAP", 256);
   static const int iters = ft_env_int("FT_BNH_RED_ITERS", 8);
   long b = (tasks + FT_BLOCK * iters - 1) / (FT_BLOCK * iters);
   if (b > cap) b = cap;
@@ -992,7 +992,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
       const long NI = N * HW, tasks = NI << lgc;
       const int B = bnh_red_grid(tasks);
       auto part = torch::empty({B, C, 2}, f32);
-      if (relu && VN == 8)  // ReLU bitmask: bwd reads 1 bit/elem, not y
+      static const bool use_mask = ft_env_int("FT_BNH_MASK", 1) != 0;
+      if (relu && VN == 8 && use_mask)  // bwd reads 1 bit/elem instead of y
         mask = torch::empty({(NI * C) >> 3},
                             x.options().dtype(torch::kByte));
       hipLaunchKernelGGL((bnh_stats_k<T, typename BnVec<T>::V, VN>),
