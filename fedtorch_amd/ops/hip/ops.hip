@@ -992,7 +992,10 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
       const long NI = N * HW, tasks = NI << lgc;
       const int B = bnh_red_grid(tasks);
       auto part = torch::empty({B, C, 2}, f32);
-      static const bool use_mask = ft_env_int("FT_BNH_MASK", 1) != 0;
+      // measured ~1-2% SLOWER at ResNet-20/b256 (y re-reads are
+      // LLC-hot; byte traffic adds overhead) — default off, kept for
+      // larger feature maps (profiles/r01_bench_notes.md)
+      static const bool use_mask = ft_env_int("FT_BNH_MASK", 0) != 0;
       if (relu && VN == 8 && use_mask)  // bwd reads 1 bit/elem instead of y
         mask = torch::empty({(NI * C) >> 3},
                             x.options().dtype(torch::kByte));
