@@ -235,3 +235,26 @@ def test_arena_bf16_compute_cpu_roundtrip():
     assert m[0].weight.dtype == torch.float32
     assert torch.allclose(a.flat, ref * 2.0)
     assert a.check_views()
+
+
+def test_stem_conversion_gating():
+    """convert_stem swaps only 3x3/s1/p1 Ci=3 Co in {16,32} stems; ImageNet
+    7x7 stems and densenet growth stems stay stock Conv2d."""
+    from types import SimpleNamespace
+    import torch.nn as nn
+    from fedtorch_amd.components.model import define_model
+    from fedtorch_amd.ops.stemconv import convert_stem, NhwcStemConv
+
+    a = SimpleNamespace(arch='resnet20', data='cifar10', debug=False)
+    m = convert_stem(define_model(a))
+    assert isinstance(m.conv1, NhwcStemConv)
+
+    a = SimpleNamespace(arch='resnet18', data='imagenet', debug=False)
+    m = convert_stem(define_model(a))
+    assert type(m.conv1) is nn.Conv2d  # 7x7/s2 stem: not converted
+
+    a = SimpleNamespace(arch='densenet121', data='cifar10', debug=False,
+                        densenet_growth_rate=12, densenet_bc_mode=False,
+                        densenet_compression=0.5, drop_rate=0.0)
+    m = convert_stem(define_model(a))
+    assert type(m.conv1) is nn.Conv2d  # Co=24: not a compiled shape
