@@ -28,7 +28,6 @@ sys.path.insert(0, REPO)
 from fedtorch_amd.parameters import get_args  # noqa: E402
 from fedtorch_amd.nodes import Client  # noqa: E402
 from fedtorch_amd.trainings.federated import amp  # noqa: E402
-from fedtorch_amd.trainings.eval import inference  # noqa: E402
 from fedtorch_amd.aggregation.federated import (  # noqa: E402
     fedavg_aggregation, fedgate_aggregation, aggregate_bn_buffers)
 
