@@ -27,8 +27,11 @@ def _num_classes(dataset):
 
 
 def conv3x3(in_planes, planes, stride=1):
-    return nn.Conv2d(in_planes, planes, kernel_size=3, stride=stride,
-                     padding=1, bias=False)
+    # NhwcConv3x3 = stock Conv2d whose channels_last bf16 wrw runs the MFMA
+    # kernel (ops/conv3x3.py); every other configuration falls back inline.
+    from fedtorch_amd.ops.conv3x3 import NhwcConv3x3
+    return NhwcConv3x3(in_planes, planes, kernel_size=3, stride=stride,
+                       padding=1, bias=False)
 
 
 class BasicBlock(nn.Module):

@@ -14,6 +14,7 @@
 #include "common.h"
 #include "batchnorm.h"
 #include "stemconv.h"
+#include "convwrw.h"
 
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 #define STREAM at::hip::getCurrentHIPStream().stream()
@@ -808,6 +809,86 @@ void cast_to_half(torch::Tensor src, torch::Tensor dst) {
 }
 
 // ==========================================================================
+// MFMA 3x3/s1/p1 NHWC bf16 conv weight gradient (convwrw.h)
+// ==========================================================================
+torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && dy.dim() == 4 && x.dim() == 4,
+              "conv3x3_wrw: 4D GPU tensors");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast)
+                  && x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3_wrw: channels_last only");
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16
+                  && x.scalar_type() == torch::kBFloat16,
+              "conv3x3_wrw: bf16 only");
+  const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = dy.size(1);
+  TORCH_CHECK(dy.size(0) == N && dy.size(2) == H && dy.size(3) == W,
+              "conv3x3_wrw: stride-1 same-shape only");
+  const bool ok = (Co == Ci) && ((Co == 16 && W == 32) ||
+                                 (Co == 32 && W == 16) ||
+                                 (Co == 64 && W == 8));
+  TORCH_CHECK(ok, "conv3x3_wrw: unsupported (Co,Ci,W)=", Co, ",", Ci, ",", W);
+  const int R = 32 / W;
+  TORCH_CHECK(H % R == 0, "conv3x3_wrw: H % (32/W) != 0");
+  const long tiles = (long)N * (H / R);
+  auto f32 = x.options().dtype(torch::kFloat);
+  int nblk, groups;
+  if (Co == 16) {  // Q==1: 4 tile streams per block
+    groups = 4;
+    long b = (tiles + 4 * 8 - 1) / (4 * 8);  // >=8 tiles per stream
+    nblk = (int)(b < 1 ? 1 : (b > 256 ? 256 : b));
+  } else {
+    groups = 1;
+    long b = tiles > 512 ? 512 : tiles;
+    nblk = (int)(b < 1 ? 1 : b);
+  }
+  const long rows = (long)nblk * groups;
+  auto part = torch::empty({rows, 9, (long)Co, (long)Ci}, f32);
+  auto dw = torch::empty(
+      {Co, Ci, 3, 3},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const __hip_bfloat16* dyp =
+      reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr());
+  const __hip_bfloat16* xp =
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());
+  if (Co == 16)
+    hipLaunchKernelGGL((conv3x3_wrw_k<16, 16, 32>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H);
+  else if (Co == 32)
+    hipLaunchKernelGGL((conv3x3_wrw_k<32, 32, 16>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H);
+  else
+    hipLaunchKernelGGL((conv3x3_wrw_k<64, 64, 8>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H);
+  const int wn = Co * 9 * Ci;
+  hipLaunchKernelGGL(conv3x3_wrw_final_k,
+                     dim3((wn + FT_BLOCK / WAVE - 1) / (FT_BLOCK / WAVE)),
+                     dim3(FT_BLOCK), 0, STREAM, part.data_ptr<float>(),
+                     rows, wn,
+                     reinterpret_cast<__hip_bfloat16*>(dw.data_ptr()), Co,
+                     Ci);
+  return dw;
+}
+
+// ==========================================================================
+// MFMA layout probe (convwrw.h) — used by the GPU layout test
+// ==========================================================================
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  CHK(A); CHK(B);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32
+                  && B.numel() == 32 * 16, "probe expects A[16,32] B[32,16] bf16");
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(mfma_probe_gemm, dim3(1), dim3(64), 0, STREAM,
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),
+                     D.data_ptr<float>());
+  return D;
+}
+
+// ==========================================================================
 // NHWC stem convolution (kernels in stemconv.h)
 // ==========================================================================
 static inline bool stem_cl(const torch::Tensor& t) {
@@ -1177,4 +1258,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stem_conv_fwd", &stem_conv_fwd);
   m.def("stem_conv_wrw", &stem_conv_wrw);
   m.def("cast_to_half", &cast_to_half);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("conv3x3_wrw", &conv3x3_wrw);
 }

@@ -63,7 +63,7 @@ def convert_stem(model):
     """Replace an eligible `conv1` stem with NhwcStemConv (weights reused in
     place; call BEFORE Arena construction)."""
     conv = getattr(model, 'conv1', None)
-    if (isinstance(conv, nn.Conv2d) and type(conv) is nn.Conv2d
+    if (isinstance(conv, nn.Conv2d) and not isinstance(conv, NhwcStemConv)
             and conv.kernel_size == (3, 3) and conv.stride == (1, 1)
             and conv.padding == (1, 1) and conv.bias is None
             and conv.in_channels == 3 and conv.out_channels in (16, 32)):
