@@ -38,8 +38,9 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 //             each owns one 16x16 quadrant of the 32x32 output.
 //   Q == 16 : like Q == 4 with 4 quadrant passes (K re-read; the whole
 //             dY/X working set is LLC-resident at these sizes).
-// Partials land in part[rows][9][CO][CI] fp32; wrw_final_k reduces rows
-// and emits bf16 dW in channels_last memory order [co][kh][kw][ci].
+// Partials land rows-contiguous in part[9*CO*CI][rows] fp32 (so the
+// final reduction reads coalesced); wrw_final_k reduces rows and emits
+// bf16 dW in channels_last memory order [co][kh][kw][ci].
 // ==========================================================================
 template <int CO, int CI, int W>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
@@ -84,21 +85,34 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
     for (long tg = group_id0; tg < tiles; tg += gstride) {
       const int n = (int)(tg / rows_per_img);
       const int h0 = (int)(tg % rows_per_img) * R;
-      // ---- stage dY tile [32 positions][CO] --------------------------------
+      // ---- stage dY tile [32 positions][CO], 16 B chunks ------------------
       const __hip_bfloat16* dyp =
           dy + (((long)n * H + h0) * W) * CO;
-      for (int e = gtid; e < 32 * CO; e += gthreads)
-        s_dy[grp][e] = dyp[e];
+      constexpr int DYC = 32 * CO / 8;  // 16 B chunks
+      for (int e = gtid; e < DYC; e += gthreads)
+        reinterpret_cast<uint4*>(&s_dy[grp][0])[e] =
+            reinterpret_cast<const uint4*>(dyp)[e];
       // ---- stage X slab rows h0-1..h0+R with zero halo ring ---------------
-      for (int e = gtid; e < XROWS * XCOLS * CI; e += gthreads) {
-        const int ci = e % CI;
-        const int col = (e / CI) % XCOLS;
-        const int row = e / (CI * XCOLS);
-        const int hh = h0 - 1 + row, ww = col - 1;
-        __hip_bfloat16 v = (__hip_bfloat16)0.f;
-        if (hh >= 0 && hh < H && ww >= 0 && ww < W)
-          v = x[(((long)n * H + hh) * W + ww) * CI + ci];
-        s_x[grp][e] = v;
+      // halo ring (cols 0 and W+1 of every row + fully-OOB rows) zeroed,
+      // then valid row bodies copied in 16 B chunks
+      constexpr int HCH = CI / 8;        // 16 B chunks per halo column
+      for (int e = gtid; e < XROWS * 2 * HCH; e += gthreads) {
+        const int row = e / (2 * HCH), half = (e / HCH) & 1,
+                  c = e % HCH;
+        const int col = half ? (W + 1) : 0;
+        reinterpret_cast<uint4*>(
+            &s_x[grp][(row * XCOLS + col) * CI])[c] = uint4{0, 0, 0, 0};
+      }
+      constexpr int BCH = W * CI / 8;    // 16 B chunks per row body
+      for (int e = gtid; e < XROWS * BCH; e += gthreads) {
+        const int row = e / BCH, c = e % BCH;
+        const int hh = h0 - 1 + row;
+        uint4 v = {0, 0, 0, 0};
+        if (hh >= 0 && hh < H)
+          v = reinterpret_cast<const uint4*>(
+              x + (((long)n * H + hh) * W) * CI)[c];
+        reinterpret_cast<uint4*>(
+            &s_x[grp][(row * XCOLS + 1) * CI])[c] = v;
       }
       if (GW > 1) __syncthreads();
       // ---- fragments + 9 MFMA ---------------------------------------------
@@ -123,22 +137,24 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
       }
       if (GW > 1) __syncthreads();  // before the next stage overwrites
     }
-    // ---- write partials: row = tile-stream id -----------------------------
+    // ---- write partials (rows-contiguous: [9*CO*CI][nrows]) --------------
     const long out_row = (long)blockIdx.x * GROUPS + grp;
-    float* pr = part + ((out_row * 9) * CO) * CI;
+    const long nrows = (long)gridDim.x * GROUPS;
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
       const int row = (lane >> 4) * 4;  // + reg
       const int col = lane & 15;
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        pr[((long)t * CO + qco + row + r) * CI + qci + col] = acc[t][r];
+        part[(((long)t * CO + qco + row + r) * CI + qci + col) * nrows
+             + out_row] = acc[t][r];
     }
   }
 }
 
-// reduce part[rows][9][CO][CI] -> dw bf16 [co][kh][kw][ci] (channels_last
-// conv-weight memory order); one wave per output element set.
+// reduce part[9*CO*CI][rows] -> dw bf16 [co][kh][kw][ci] (channels_last
+// conv-weight memory order); one wave per output element, contiguous
+// coalesced row reads + xor-butterfly.
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_final_k(
     const float* __restrict__ part, long rows, int wn,
     __hip_bfloat16* __restrict__ dw, int CO, int CI) {
@@ -149,10 +165,9 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_final_k(
   const int ci = t % CI;
   const int tap = (t / CI) % 9;
   const int co = t / (9 * CI);
-  const long off = ((long)tap * CO + co) * CI + ci;
-  const long rstride = (long)9 * CO * CI;
+  const float* p = part + (((long)tap * CO + co) * CI + ci) * rows;
   float s = 0.f;
-  for (long r = lane; r < rows; r += WAVE) s += part[r * rstride + off];
+  for (long r = lane; r < rows; r += WAVE) s += p[r];
 #pragma unroll
   for (int o = WAVE / 2; o; o >>= 1) s += __shfl_xor(s, o, WAVE);
   if (lane == 0) dw[t] = __float2bfloat16(s);

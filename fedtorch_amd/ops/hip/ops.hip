@@ -833,17 +833,17 @@ torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
   const long tiles = (long)N * (H / R);
   auto f32 = x.options().dtype(torch::kFloat);
   int nblk, groups;
-  if (Co == 16) {  // Q==1: 4 tile streams per block
+  if (Co == 16) {  // Q==1: 4 independent tile streams per block
     groups = 4;
-    long b = (tiles + 4 * 8 - 1) / (4 * 8);  // >=8 tiles per stream
-    nblk = (int)(b < 1 ? 1 : (b > 256 ? 256 : b));
+    long b = (tiles + 4 * 4 - 1) / (4 * 4);  // >=4 tiles per stream
+    nblk = (int)(b < 1 ? 1 : (b > 512 ? 512 : b));
   } else {
     groups = 1;
     long b = tiles > 512 ? 512 : tiles;
     nblk = (int)(b < 1 ? 1 : b);
   }
   const long rows = (long)nblk * groups;
-  auto part = torch::empty({rows, 9, (long)Co, (long)Ci}, f32);
+  auto part = torch::empty({(long)9 * Co * Ci, rows}, f32);
   auto dw = torch::empty(
       {Co, Ci, 3, 3},
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
