@@ -832,14 +832,16 @@ torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(H % R == 0, "conv3x3_wrw: H % (32/W) != 0");
   const long tiles = (long)N * (H / R);
   auto f32 = x.options().dtype(torch::kFloat);
+  static const int cap1 = ft_env_int("FT_WRW_NBLK1", 512);
+  static const int capq = ft_env_int("FT_WRW_NBLKQ", 512);
   int nblk, groups;
   if (Co == 16) {  // Q==1: 4 independent tile streams per block
     groups = 4;
     long b = (tiles + 4 * 4 - 1) / (4 * 4);  // >=4 tiles per stream
-    nblk = (int)(b < 1 ? 1 : (b > 512 ? 512 : b));
+    nblk = (int)(b < 1 ? 1 : (b > cap1 ? cap1 : b));
   } else {
     groups = 1;
-    long b = tiles > 512 ? 512 : tiles;
+    long b = tiles > capq ? capq : tiles;
     nblk = (int)(b < 1 ? 1 : b);
   }
   const long rows = (long)nblk * groups;
