@@ -19,6 +19,13 @@
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 #define STREAM at::hip::getCurrentHIPStream().stream()
 
+// grid caps overridable via env for on-box tuning sweeps (read once)
+static inline int ft_env_int(const char* k, int d) {
+  const char* v = getenv(k);
+  return v ? atoi(v) : d;
+}
+
+
 // ==========================================================================
 // fused dual-mode SGD step (+ per-algorithm corrections)
 // ==========================================================================
@@ -1025,11 +1032,6 @@ static inline int bnh_lgc(long C, int VN) {
   int l = 0;
   for (long c = C / VN; c > 1; c >>= 1) ++l;
   return l;
-}
-// grid caps overridable via env for on-box tuning sweeps (read once)
-static inline int ft_env_int(const char* k, int d) {
-  const char* v = getenv(k);
-  return v ? atoi(v) : d;
 }
 static inline int bnh_red_grid(long tasks) {
   static const int cap = ft_env_int("FT_BNH_RED_CAP", 256);
