@@ -38,9 +38,10 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 //             each owns one 16x16 quadrant of the 32x32 output.
 //   Q == 16 : like Q == 4 with 4 quadrant passes (K re-read; the whole
 //             dY/X working set is LLC-resident at these sizes).
-// Partials land rows-contiguous in part[9*CO*CI][rows] fp32 (so the
-// final reduction reads coalesced); wrw_final_k reduces rows and emits
-// bf16 dW in channels_last memory order [co][kh][kw][ci].
+// Partials land row-major in part[rows][9*CO*CI] fp32 (writes coalesce:
+// 16 consecutive-ci lanes store 64 B bursts); wrw_final_k sums the rows
+// of 64-column tiles (coalesced reads) and emits bf16 dW in channels_last
+// memory order [co][kh][kw][ci].
 // ==========================================================================
 template <int CO, int CI, int W>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
@@ -137,40 +138,44 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
       }
       if (GW > 1) __syncthreads();  // before the next stage overwrites
     }
-    // ---- write partials (rows-contiguous: [9*CO*CI][nrows]) --------------
+    // ---- write partials (row-major [rows][9*CO*CI]: coalesced) -----------
     const long out_row = (long)blockIdx.x * GROUPS + grp;
-    const long nrows = (long)gridDim.x * GROUPS;
+    float* pr = part + out_row * (long)(9 * CO * CI);
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
       const int row = (lane >> 4) * 4;  // + reg
       const int col = lane & 15;
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        part[(((long)t * CO + qco + row + r) * CI + qci + col) * nrows
-             + out_row] = acc[t][r];
+        pr[((long)t * CO + qco + row + r) * CI + qci + col] = acc[t][r];
     }
   }
 }
 
-// reduce part[9*CO*CI][rows] -> dw bf16 [co][kh][kw][ci] (channels_last
-// conv-weight memory order); one wave per output element, contiguous
-// coalesced row reads + xor-butterfly.
+// reduce part[rows][9*CO*CI] -> dw bf16 [co][kh][kw][ci] (channels_last
+// conv-weight memory order).  Each block owns 64 consecutive part-columns
+// and walks the rows with 4 row-threads per column: every row visit is a
+// 256 B contiguous read, so the reduction streams the partial buffer at
+// full bandwidth instead of one 4 B element per 64 B line.
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_final_k(
     const float* __restrict__ part, long rows, int wn,
     __hip_bfloat16* __restrict__ dw, int CO, int CI) {
-  const int wave = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
-  const int t = blockIdx.x * (FT_BLOCK / WAVE) + wave;
-  if (t >= wn) return;
-  // t indexes [co][kh][kw][ci] memory order; map to part's [tap][co][ci]
-  const int ci = t % CI;
-  const int tap = (t / CI) % 9;
-  const int co = t / (9 * CI);
-  const float* p = part + (((long)tap * CO + co) * CI + ci) * rows;
+  __shared__ float lds[4][64];
+  const int c = threadIdx.x & 63, rs = threadIdx.x >> 6;
+  const long off = (long)blockIdx.x * 64 + c;
   float s = 0.f;
-  for (long r = lane; r < rows; r += WAVE) s += p[r];
-#pragma unroll
-  for (int o = WAVE / 2; o; o >>= 1) s += __shfl_xor(s, o, WAVE);
-  if (lane == 0) dw[t] = __float2bfloat16(s);
+  if (off < wn)
+    for (long r = rs; r < rows; r += 4) s += part[r * wn + off];
+  lds[rs][c] = s;
+  __syncthreads();
+  if (rs == 0 && off < wn) {
+    s = lds[0][c] + lds[1][c] + lds[2][c] + lds[3][c];
+    // part column (tap, co, ci) -> dw memory index (co, tap, ci)
+    const int ci = (int)(off % CI);
+    const int co = (int)((off / CI) % CO);
+    const int tap = (int)(off / ((long)CO * CI));
+    dw[((long)co * 9 + tap) * CI + ci] = __float2bfloat16(s);
+  }
 }
 
 // ---- layout probe: D[16,16] = A[16,32] x B[32,16], one wave ---------------

@@ -839,8 +839,8 @@ torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(H % R == 0, "conv3x3_wrw: H % (32/W) != 0");
   const long tiles = (long)N * (H / R);
   auto f32 = x.options().dtype(torch::kFloat);
-  static const int cap1 = ft_env_int("FT_WRW_NBLK1", 512);
-  static const int capq = ft_env_int("FT_WRW_NBLKQ", 512);
+  static const int cap1 = ft_env_int("FT_WRW_NBLK1", 256);
+  static const int capq = ft_env_int("FT_WRW_NBLKQ", 128);
   int nblk, groups;
   if (Co == 16) {  // Q==1: 4 independent tile streams per block
     groups = 4;
@@ -852,7 +852,7 @@ torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
     nblk = (int)(b < 1 ? 1 : b);
   }
   const long rows = (long)nblk * groups;
-  auto part = torch::empty({(long)9 * Co * Ci, rows}, f32);
+  auto part = torch::empty({rows, (long)9 * Co * Ci}, f32);
   auto dw = torch::empty(
       {Co, Ci, 3, 3},
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
@@ -873,8 +873,7 @@ torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
                        dim3(FT_BLOCK), 0, STREAM, dyp, xp,
                        part.data_ptr<float>(), N, H);
   const int wn = Co * 9 * Ci;
-  hipLaunchKernelGGL(conv3x3_wrw_final_k,
-                     dim3((wn + FT_BLOCK / WAVE - 1) / (FT_BLOCK / WAVE)),
+  hipLaunchKernelGGL(conv3x3_wrw_final_k, dim3((wn + 63) / 64),
                      dim3(FT_BLOCK), 0, STREAM, part.data_ptr<float>(),
                      rows, wn,
                      reinterpret_cast<__hip_bfloat16*>(dw.data_ptr()), Co,
