@@ -12,6 +12,8 @@ Eligible shapes (compiled template instances): Co == Ci with
 (Co, W) in {(16,32), (32,16), (64,8)} — the CIFAR ResNet body convs.
 Anything else falls back to stock F.conv2d autograd.
 """
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -20,6 +22,12 @@ from fedtorch_amd import ops
 
 _CL = torch.channels_last
 _SHAPES = {(16, 32), (32, 16), (64, 8)}
+# Default OFF: the kernel is numerically exact (test_conv3x3_wrw_*) but at
+# CIFAR sizes it measured 27-79 us/call vs MIOpen's 22-33 us (the LDS-staged
+# tile pipeline is latency-bound at these tiny K-tiles; see
+# profiles/r01_bench_notes.md "MFMA wrw" entry).  FEDTORCH_MFMA_WRW=1
+# re-enables for experimentation.
+_ENABLED = os.environ.get('FEDTORCH_MFMA_WRW', '0') == '1'
 
 
 class _Conv3x3Fn(torch.autograd.Function):
@@ -48,7 +56,8 @@ class NhwcConv3x3(nn.Conv2d):
 
     def forward(self, x):
         w = self.weight
-        use = (torch.is_grad_enabled() and self.training and x.is_cuda
+        use = (_ENABLED and torch.is_grad_enabled() and self.training
+               and x.is_cuda
                and x.dim() == 4 and self.bias is None
                and self.stride == (1, 1) and self.padding == (1, 1)
                and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16
