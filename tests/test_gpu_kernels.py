@@ -599,7 +599,9 @@ def test_conv3x3_wrw_matches_torch(C, W):
 def test_nhwc_conv3x3_module_grads():
     """NhwcConv3x3 end-to-end: dx and dw match the stock conv autograd."""
     import torch.nn.functional as F
+    import fedtorch_amd.ops.conv3x3 as c3
     from fedtorch_amd.ops.conv3x3 import NhwcConv3x3
+    c3._ENABLED = True  # custom wrw path is env-gated off by default
     torch.manual_seed(42)
     cl = torch.channels_last
     N, C, H, W = 8, 16, 32, 32
