@@ -121,3 +121,27 @@ def test_tools_parse_roundtrip(tmp_path):
     assert len(df) == 2
     assert df['top1'].tolist() == [55.0, 60.0]
     assert df['time'].tolist() == [0.0, 5.0]
+
+
+def test_bench_two_rank_cpu_contract():
+    """The driver launches bench.py via torchrun with N ranks; verify the
+    multi-rank path (gloo on CPU) emits the contract JSON from rank 0."""
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR='127.0.0.1', MASTER_PORT='29881')
+    out = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29882', os.path.join(repo, 'bench.py'),
+         '--gpus', '2', '--steps', '12', '--warmup', '2', '--batch', '16'],
+        capture_output=True, text=True, timeout=600, cwd=repo, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith('{')]
+    assert len(lines) == 1, lines  # rank 0 only
+    d = json.loads(lines[0])
+    assert d['n_gpus'] == 2
+    assert d['config']['global_batch'] == 32
+    assert d['value'] > 0
