@@ -55,14 +55,18 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
   constexpr int PASSES = (Q + 3) / 4;   // quadrant passes per tile stream
   constexpr int XROWS = R + 2, XCOLS = W + 2;
 
-  __shared__ __hip_bfloat16 s_dy[GROUPS][32 * CO];
-  __shared__ __hip_bfloat16 s_x[GROUPS][XROWS * XCOLS * CI];
+  // double-buffered slabs: tile t+1's global loads are issued (into
+  // registers) while tile t's fragments/MFMAs run, so the waves are not
+  // parked on the load->ds_write->ds_read chain every tile (PMC: the
+  // single-buffer version spent 68 % of wave cycles in SQ_WAIT_ANY).
+  __shared__ __hip_bfloat16 s_dy[GROUPS][2][32 * CO];
+  __shared__ __hip_bfloat16 s_x[GROUPS][2][XROWS * XCOLS * CI];
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int grp = (GW == 1) ? wave : 0;
   const int gtid = (GW == 1) ? lane : threadIdx.x;  // tid within the group
-  const int gthreads = GW * WAVE;
+  constexpr int GTH = GW * WAVE;
 
   const int rows_per_img = H / R;
   const long tiles = (long)N * rows_per_img;
@@ -75,6 +79,63 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
   const int kbase = (lane >> 4) * 8;
   const int r0 = kbase / W, w0 = kbase % W;
 
+  constexpr int DYC = 32 * CO / 8;   // 16 B chunks of the dY tile
+  constexpr int BCH = W * CI / 8;    // 16 B chunks per X row body
+  constexpr int HCH = CI / 8;        // 16 B chunks per halo column
+  constexpr int DYI = (DYC + GTH - 1) / GTH;
+  constexpr int XBI = (XROWS * BCH + GTH - 1) / GTH;
+
+  uint4 dyr[DYI], xr[XBI];
+
+  auto load_tile = [&](long tg) {
+    const int n = (int)(tg / rows_per_img);
+    const int h0 = (int)(tg % rows_per_img) * R;
+    const __hip_bfloat16* dyp = dy + (((long)n * H + h0) * W) * CO;
+#pragma unroll
+    for (int j = 0; j < DYI; ++j) {
+      const int e = gtid + j * GTH;
+      if (e < DYC)
+        dyr[j] = reinterpret_cast<const uint4*>(dyp)[e];
+    }
+#pragma unroll
+    for (int j = 0; j < XBI; ++j) {
+      const int e = gtid + j * GTH;
+      uint4 v = {0, 0, 0, 0};
+      if (e < XROWS * BCH) {
+        const int row = e / BCH, c = e % BCH;
+        const int hh = h0 - 1 + row;
+        if (hh >= 0 && hh < H)
+          v = reinterpret_cast<const uint4*>(
+              x + (((long)n * H + hh) * W) * CI)[c];
+      }
+      xr[j] = v;
+    }
+  };
+
+  auto write_slab = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < DYI; ++j) {
+      const int e = gtid + j * GTH;
+      if (e < DYC)
+        reinterpret_cast<uint4*>(&s_dy[grp][buf][0])[e] = dyr[j];
+    }
+    for (int e = gtid; e < XROWS * 2 * HCH; e += GTH) {
+      const int row = e / (2 * HCH), half = (e / HCH) & 1, c = e % HCH;
+      const int col = half ? (W + 1) : 0;
+      reinterpret_cast<uint4*>(
+          &s_x[grp][buf][(row * XCOLS + col) * CI])[c] = uint4{0, 0, 0, 0};
+    }
+#pragma unroll
+    for (int j = 0; j < XBI; ++j) {
+      const int e = gtid + j * GTH;
+      if (e < XROWS * BCH) {
+        const int row = e / BCH, c = e % BCH;
+        reinterpret_cast<uint4*>(
+            &s_x[grp][buf][(row * XCOLS + 1) * CI])[c] = xr[j];
+      }
+    }
+  };
+
   for (int pass = 0; pass < PASSES; ++pass) {
     const int q = pass * 4 + ((GW == 1) ? 0 : wave);
     const int qco = (Q == 1) ? 0 : (q / QC) * 16;
@@ -83,45 +144,21 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
 #pragma unroll
     for (int t = 0; t < 9; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
 
+    int cur = 0;
+    if (group_id0 < tiles) {
+      load_tile(group_id0);
+      write_slab(cur);
+    }
+    if (GW > 1) __syncthreads();
     for (long tg = group_id0; tg < tiles; tg += gstride) {
-      const int n = (int)(tg / rows_per_img);
-      const int h0 = (int)(tg % rows_per_img) * R;
-      // ---- stage dY tile [32 positions][CO], 16 B chunks ------------------
-      const __hip_bfloat16* dyp =
-          dy + (((long)n * H + h0) * W) * CO;
-      constexpr int DYC = 32 * CO / 8;  // 16 B chunks
-      for (int e = gtid; e < DYC; e += gthreads)
-        reinterpret_cast<uint4*>(&s_dy[grp][0])[e] =
-            reinterpret_cast<const uint4*>(dyp)[e];
-      // ---- stage X slab rows h0-1..h0+R with zero halo ring ---------------
-      // halo ring (cols 0 and W+1 of every row + fully-OOB rows) zeroed,
-      // then valid row bodies copied in 16 B chunks
-      constexpr int HCH = CI / 8;        // 16 B chunks per halo column
-      for (int e = gtid; e < XROWS * 2 * HCH; e += gthreads) {
-        const int row = e / (2 * HCH), half = (e / HCH) & 1,
-                  c = e % HCH;
-        const int col = half ? (W + 1) : 0;
-        reinterpret_cast<uint4*>(
-            &s_x[grp][(row * XCOLS + col) * CI])[c] = uint4{0, 0, 0, 0};
-      }
-      constexpr int BCH = W * CI / 8;    // 16 B chunks per row body
-      for (int e = gtid; e < XROWS * BCH; e += gthreads) {
-        const int row = e / BCH, c = e % BCH;
-        const int hh = h0 - 1 + row;
-        uint4 v = {0, 0, 0, 0};
-        if (hh >= 0 && hh < H)
-          v = reinterpret_cast<const uint4*>(
-              x + (((long)n * H + hh) * W) * CI)[c];
-        reinterpret_cast<uint4*>(
-            &s_x[grp][(row * XCOLS + 1) * CI])[c] = v;
-      }
-      if (GW > 1) __syncthreads();
-      // ---- fragments + 9 MFMA ---------------------------------------------
+      const long nxt = tg + gstride;
+      if (nxt < tiles) load_tile(nxt);  // loads in flight during compute
+      // ---- fragments + 9 MFMA from slab `cur` -----------------------------
       bf16x8 a;
 #pragma unroll
       for (int i = 0; i < 8; ++i)
         a[i] = *reinterpret_cast<const __bf16*>(
-            &s_dy[grp][(kbase + i) * CO + qco + fm]);
+            &s_dy[grp][cur][(kbase + i) * CO + qco + fm]);
 #pragma unroll
       for (int dh = 0; dh < 3; ++dh) {
 #pragma unroll
@@ -130,13 +167,17 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw_k(
 #pragma unroll
           for (int i = 0; i < 8; ++i)
             b[i] = *reinterpret_cast<const __bf16*>(
-                &s_x[grp][((r0 + dh) * XCOLS + (w0 + dw + i)) * CI + qci
-                          + fm]);
+                &s_x[grp][cur][((r0 + dh) * XCOLS + (w0 + dw + i)) * CI
+                               + qci + fm]);
           acc[dh * 3 + dw] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a, b, acc[dh * 3 + dw], 0, 0, 0);
         }
       }
-      if (GW > 1) __syncthreads();  // before the next stage overwrites
+      if (nxt < tiles) {
+        write_slab(cur ^ 1);
+        cur ^= 1;
+      }
+      if (GW > 1) __syncthreads();
     }
     // ---- write partials (row-major [rows][9*CO*CI]: coalesced) -----------
     const long out_row = (long)blockIdx.x * GROUPS + grp;
