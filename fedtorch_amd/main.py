@@ -38,6 +38,14 @@ def init_distributed(args):
 
 def main(args):
     init_distributed(args)
+    try:
+        _dispatch(args)
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def _dispatch(args):
     client = Client(args, dist.get_rank())
     client.initialize()
     client.initialize_dataset()
