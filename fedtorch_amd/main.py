@@ -18,8 +18,11 @@ from fedtorch_amd.trainings.federated import train_and_validate_federated
 
 
 def init_distributed(args):
+    """Init the process group; returns True if THIS call created it (the
+    caller then owns teardown — an externally created group is left
+    alone)."""
     if dist.is_initialized():
-        return
+        return False
     backend = args.dist_backend
     if backend in ('mpi', None):
         backend = 'nccl' if torch.cuda.is_available() else 'gloo'
@@ -34,14 +37,15 @@ def init_distributed(args):
     os.environ.setdefault('RANK', '0')
     os.environ.setdefault('WORLD_SIZE', '1')
     dist.init_process_group(backend)
+    return True
 
 
 def main(args):
-    init_distributed(args)
+    owns_pg = init_distributed(args)
     try:
         _dispatch(args)
     finally:
-        if dist.is_initialized():
+        if owns_pg and dist.is_initialized():
             dist.destroy_process_group()
 
 
