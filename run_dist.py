@@ -22,6 +22,8 @@ MLP_SIZE = {'mnist': 200, 'fashion_mnist': 200, 'cifar10': 200,
 
 
 def main(args):
+    if getattr(args, 'tmp_dir', None):
+        os.environ['TMPDIR'] = args.tmp_dir
     blocks = str(args.num_clients)
     world = ','.join(str(x) for x in range(args.num_clients))
     params = {
@@ -135,6 +137,8 @@ if __name__ == '__main__':
     parser.add_argument('-B', '--quantized_bits', default=8, type=int)
     parser.add_argument('-pm', '--fedprox_mu', default=0.002, type=float)
     parser.add_argument('-a', '--arch', default=None, type=str)
+    # reference `run_mpi.py:140`: TMPDIR override for dataset staging
+    parser.add_argument('-td', '--tmp_dir', default='/tmp', type=str)
     parser.add_argument('--bf16', action='store_true')
     parser.add_argument('--master_port', default=29500, type=int)
     sys.exit(main(parser.parse_args()))
