@@ -71,3 +71,54 @@ def test_centered_resnet_bn_stats_aggregated():
     loss = Server.global_test_tracker['losses'].avg
     assert loss < 2.5, 'server eval loss %.2f (BN stats broken?)' % loss
     assert acc > 25.0, 'test top1 %.1f' % acc
+
+
+def test_centered_apfl_personal_learns():
+    """APFL (personalized) centered loop learns past chance on the
+    personal-blend validation (reference `centered/apfl.py`)."""
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes.centered import ClientCentered, ServerCentered
+    from fedtorch_amd.trainings.centered.apfl import (
+        train_and_validate_apfl_centered)
+    argv = ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+            '--federated_type', 'apfl', '--fed_personal', 'true',
+            '--fed_personal_alpha', '0.5', '--num_comms', '4',
+            '--online_client_rate', '1.0', '-b', '50', '--lr', '0.05',
+            '--on_cuda', 'false', '-j', '4',
+            '--checkpoint', '/tmp/ft_conv_apfl', '--debug', 'false',
+            '--manual_seed', '3']
+    args = get_args(argv)
+    args.num_workers = 4
+    Clients = {}
+    for i in range(4):
+        Clients[i] = ClientCentered(args, i) if i == 0 else \
+            ClientCentered(args, i, Partitioner=Clients[0].Partitioner)
+    Server = ServerCentered(Clients[0].args, Clients[0].model)
+    train_and_validate_apfl_centered(Clients, Server)
+    acc = Server.local_personal_val_tracker['top1'].avg
+    assert acc > 30.0, acc
+
+
+def test_centered_perfedme_learns():
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes.centered import ClientCentered, ServerCentered
+    from fedtorch_amd.trainings.centered.perfedme import (
+        train_and_validate_perfedme_centered)
+    argv = ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+            '--federated_type', 'perfedme', '--num_comms', '4',
+            '--online_client_rate', '1.0', '-b', '50', '--lr', '0.05',
+            '--on_cuda', 'false', '-j', '4',
+            '--checkpoint', '/tmp/ft_conv_pfm', '--debug', 'false',
+            '--manual_seed', '3']
+    args = get_args(argv)
+    args.num_workers = 4
+    Clients = {}
+    for i in range(4):
+        Clients[i] = ClientCentered(args, i) if i == 0 else \
+            ClientCentered(args, i, Partitioner=Clients[0].Partitioner)
+    Server = ServerCentered(Clients[0].args, Clients[0].model)
+    train_and_validate_perfedme_centered(Clients, Server)
+    acc = Server.local_personal_val_tracker['top1'].avg
+    assert acc > 25.0, acc
