@@ -122,3 +122,86 @@ def test_centered_perfedme_learns():
     train_and_validate_perfedme_centered(Clients, Server)
     acc = Server.local_personal_val_tracker['top1'].avg
     assert acc > 25.0, acc
+
+
+def _centered_clients(argv, workers=4):
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes.centered import ClientCentered, ServerCentered
+    args = get_args(argv)
+    args.num_workers = workers
+    Clients = {}
+    for i in range(workers):
+        Clients[i] = ClientCentered(args, i) if i == 0 else \
+            ClientCentered(args, i, Partitioner=Clients[0].Partitioner)
+    Server = ServerCentered(Clients[0].args, Clients[0].model)
+    return Clients, Server
+
+
+def test_centered_afl_learns():
+    """AFL centered: lambda-weighted aggregation still learns and the dual
+    variable stays on the simplex (reference `centered/afl.py`)."""
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.trainings.centered.afl import (
+        train_and_validate_afl_centered)
+    Clients, Server = _centered_clients(
+        ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+         '--federated_type', 'afl', '--num_comms', '4',
+         '--online_client_rate', '1.0', '-b', '50', '--lr', '0.05',
+         '--on_cuda', 'false', '-j', '4', '--checkpoint', '/tmp/ft_conv_afl',
+         '--debug', 'false', '--manual_seed', '3'])
+    train_and_validate_afl_centered(Clients, Server)
+    lam = Server.lambda_vector
+    assert abs(float(lam.sum()) - 1.0) < 1e-4
+    assert float(lam.min()) >= 0.0
+    acc = Server.global_test_tracker['top1'].avg
+    assert acc > 30.0, acc
+
+
+def test_centered_drfa_learns():
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.trainings.centered.drfa import (
+        train_and_validate_drfa_centered)
+    Clients, Server = _centered_clients(
+        ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+         '--federated_type', 'fedavg', '--federated_drfa', 'true',
+         '--num_comms', '4', '--online_client_rate', '0.75',
+         '--local_step', '4', '--federated_sync_type', 'local_step',
+         '-b', '50', '--lr', '0.05', '--on_cuda', 'false', '-j', '4',
+         '--checkpoint', '/tmp/ft_conv_drfa', '--debug', 'false',
+         '--manual_seed', '3'])
+    train_and_validate_drfa_centered(Clients, Server)
+    lam = Server.lambda_vector
+    assert abs(float(lam.sum()) - 1.0) < 1e-4
+    acc = Server.global_test_tracker['top1'].avg
+    assert acc > 25.0, acc
+
+
+def test_centered_qffl_learns():
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.trainings.centered.main import (
+        train_and_validate_federated_centered)
+    Clients, Server = _centered_clients(
+        ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+         '--federated_type', 'qffl', '--qffl_q', '1.0', '--num_comms', '4',
+         '--online_client_rate', '1.0', '-b', '50', '--lr', '0.1',
+         '--on_cuda', 'false', '-j', '4', '--checkpoint', '/tmp/ft_conv_qffl',
+         '--debug', 'false', '--manual_seed', '3'])
+    train_and_validate_federated_centered(Clients, Server)
+    acc = Server.global_test_tracker['top1'].avg
+    assert acc > 25.0, acc
+
+
+def test_centered_perfedavg_learns():
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '600'
+    from fedtorch_amd.trainings.centered.main import (
+        train_and_validate_federated_centered)
+    Clients, Server = _centered_clients(
+        ['-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+         '--federated_type', 'perfedavg', '--num_comms', '4',
+         '--online_client_rate', '1.0', '-b', '50', '--lr', '0.05',
+         '--on_cuda', 'false', '-j', '4',
+         '--checkpoint', '/tmp/ft_conv_pfa', '--debug', 'false',
+         '--manual_seed', '3'])
+    train_and_validate_federated_centered(Clients, Server)
+    acc = Server.local_personal_val_tracker['top1'].avg
+    assert acc > 25.0, acc
