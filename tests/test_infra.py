@@ -145,3 +145,48 @@ def test_bench_two_rank_cpu_contract():
     assert d['n_gpus'] == 2
     assert d['config']['global_batch'] == 32
     assert d['value'] > 0
+
+
+def test_accuracy_and_per_class_oracle():
+    """metrics vs hand-computed values (reference `metrics.py:50-91`)."""
+    import torch
+    from fedtorch_amd.components.metrics import accuracy, accuracy_per_class
+    logits = torch.tensor([[0.9, 0.1, 0.0],
+                           [0.1, 0.8, 0.1],
+                           [0.2, 0.7, 0.1],
+                           [0.3, 0.3, 0.4]])
+    target = torch.tensor([0, 1, 0, 2])
+    top1, top2 = accuracy(logits, target, topk=(1, 2))
+    assert abs(top1 - 75.0) < 1e-5      # 3 of 4 correct
+    assert abs(top2 - 100.0) < 1e-5     # sample 2's true class is 2nd
+    acc, count = accuracy_per_class(logits, target, torch.tensor([0, 1, 2]))
+    assert count.tolist() == [2.0, 1.0, 1.0]
+    assert abs(acc[0].item() - 50.0) < 1e-4   # one of the two 0s right
+    assert abs(acc[1].item() - 100.0) < 1e-4
+    assert abs(acc[2].item() - 100.0) < 1e-4
+    # rnn mode: argmax over flattened (seq) predictions
+    r = accuracy(logits, target, topk=(1,), rnn=True)
+    assert abs(r[0] - 75.0) < 1e-5
+
+
+def test_checkpoint_best_and_epoch_copies(tmp_path):
+    """save_to_checkpoint produces model_best + per-epoch copies exactly as
+    the reference layout (`logs/checkpoint.py:68-82`)."""
+    import os
+    import torch
+    from types import SimpleNamespace
+    from fedtorch_amd.logs.checkpoint import save_to_checkpoint
+    d = str(tmp_path / 'ck')
+    args = SimpleNamespace(save_some_models=['2'])
+    state = {'arguments': args, 'current_epoch': 1,
+             'state_dict': {'w': torch.zeros(2)}, 'best_prec1': 10.0}
+    save_to_checkpoint(state, True, d, 'checkpoint.pth.tar')
+    assert os.path.exists(os.path.join(d, 'checkpoint.pth.tar'))
+    assert os.path.exists(os.path.join(d, 'model_best.pth.tar'))
+    assert not os.path.exists(os.path.join(d, 'checkpoint_epoch_1.pth.tar'))
+    state['current_epoch'] = 2
+    save_to_checkpoint(state, False, d, 'checkpoint.pth.tar')
+    assert os.path.exists(os.path.join(d, 'checkpoint_epoch_2.pth.tar'))
+    state['current_epoch'] = 3
+    save_to_checkpoint(state, False, d, 'checkpoint.pth.tar', save_all=True)
+    assert os.path.exists(os.path.join(d, 'checkpoint_epoch_3.pth.tar'))
