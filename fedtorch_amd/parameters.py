@@ -200,7 +200,13 @@ def build_parser():
                              'the arena hot paths (GPU only). When False, '
                              'fall back to eager torch ops.')
     parser.add_argument('--hip_graph', type=str2bool, default=False,
-                        help='capture the local-step inner loop in a hipGraph.')
+                        help='reserved. hipGraph capture of the local step '
+                             'is implemented in the benchmark path '
+                             '(bench.py --graph, two-capture stolen-grad '
+                             'scheme); the parity training loops run eager '
+                             'because per-step LR scheduling and per-step '
+                             'console metrics are part of the reference '
+                             'semantics a captured step would bake in.')
     parser.add_argument('--clients_per_rank', default=1, type=int,
                         help='virtual clients packed per GPU rank (their '
                              'replicas and aux state stay resident in HBM3E).')
