@@ -159,7 +159,7 @@ def main():
     if use_graph:
         try:
             arena = client.arena
-            if b.bf16_weights == 'on':
+            if b.bf16_weights == 'on' and use_bf16:
                 # conv/linear params live in a bf16 twin arena: no autocast
                 # weight-cast kernels per step; the fused SGD refreshes the
                 # twin and the grad gather casts bf16 grads back to fp32.
@@ -348,7 +348,7 @@ def run_packed_bench(b, client, args, xs, ys, pool_n, world, rank, on_gpu):
                        apply_out_momentum=False)
 
     if on_gpu and b.graph != 'off':
-        if b.bf16_weights == 'on':
+        if b.bf16_weights == 'on' and args.bf16:
             for sl in slots:
                 sl['arena'].enable_bf16_compute()
         side = torch.cuda.Stream()
