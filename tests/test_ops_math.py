@@ -156,3 +156,56 @@ def test_blend_and_alpha_grad():
     ga = ops.alpha_grad(lf, pf, lg, pg, 0.4)
     expected = torch.dot(pf - lf, 0.4 * pg + 0.6 * lg).item() + 0.02 * 0.4
     assert abs(ga - expected) < 1e-4
+
+
+def test_rank_weight_semantics():
+    """The weighting that makes the all-reduce identical to the reference's
+    star gather+sum+broadcast (`fedavg.py:17-27`, `qsparse.py:23`)."""
+    from types import SimpleNamespace
+    from fedtorch_amd.aggregation.federated import rank_weight
+
+    def args_for(rank, n=4):
+        return SimpleNamespace(graph=SimpleNamespace(rank=rank, n_nodes=n),
+                               num_samples_per_epoch=100,
+                               train_dataset_size=400)
+
+    # online clients include the server: uniform 1/K
+    w, k, part = rank_weight(args_for(1), [0, 1, 2])
+    assert (w, k, part) == (1.0 / 3, 3, True)
+    # server offline: counted in K but contributes 0 (`fedavg.py:19-20`)
+    w, k, part = rank_weight(args_for(0), [1, 2])
+    assert (w, k, part) == (0.0, 3, False)
+    w, k, part = rank_weight(args_for(1), [1, 2])
+    assert (w, k, part) == (1.0 / 3, 3, True)
+    # fully offline rank never contributes
+    w, _, part = rank_weight(args_for(3), [1, 2])
+    assert (w, part) == (0.0, False)
+    # DRFA/AFL: lambda_i * n / K (`fedavg.py:27`)
+    w, _, _ = rank_weight(args_for(2), [0, 1, 2], lambda_weight=0.5)
+    assert abs(w - 0.5 * 4 / 3) < 1e-9
+    # qsparse: sample-proportional (`qsparse.py:23`)
+    w, _, _ = rank_weight(args_for(2), [0, 1, 2], sample_proportional=True)
+    assert abs(w - 0.25) < 1e-9
+
+
+def test_fedadam_normalize_math():
+    """FedAdam server normalizer (the reference's `fedavg.py:81-85` crashes
+    on a missing np import; this pins our working version)."""
+    import numpy as np
+    from types import SimpleNamespace
+    import torch.nn as nn
+    from fedtorch_amd.parallel.arena import Arena
+    from fedtorch_amd.aggregation.federated import _fedadam_normalize
+    m = nn.Linear(4, 2)
+    a = Arena(m)
+    agg = torch.ones_like(a.flat)
+    args = SimpleNamespace(fedadam_beta=0.9, fedadam_tau=0.1,
+                           fedadam_v=[1.0, 1.0])
+    views = a.views_of(agg)
+    norms = [float(torch.norm(v)) for v in views]
+    _fedadam_normalize(args, a, agg)
+    for i, v in enumerate(a.views_of(agg)):
+        v_exp = 0.9 * 1.0 + 0.1 * norms[i]
+        assert abs(args.fedadam_v[i] - v_exp) < 1e-6
+        assert torch.allclose(v, torch.ones_like(v) /
+                              (np.sqrt(v_exp) + 0.1), atol=1e-6)
