@@ -209,3 +209,18 @@ def test_fedadam_normalize_math():
         assert abs(args.fedadam_v[i] - v_exp) < 1e-6
         assert torch.allclose(v, torch.ones_like(v) /
                               (np.sqrt(v_exp) + 0.1), atol=1e-6)
+
+
+def test_inference_personal_blend():
+    """alpha-blend of two models' logits (reference `eval.py:31-39`)."""
+    import torch.nn as nn
+    from fedtorch_amd.trainings.eval import inference_personal
+    torch.manual_seed(5)
+    m1, m2 = nn.Linear(6, 3), nn.Linear(6, 3)
+    x = torch.randn(10, 6)
+    y = torch.randint(0, 3, (10,))
+    crit = nn.CrossEntropyLoss()
+    loss, perf = inference_personal(m1, m2, 0.3, crit, (1,), x, y)
+    expected = crit(0.3 * m1(x) + 0.7 * m2(x), y)
+    assert torch.allclose(loss, expected, atol=1e-6)
+    assert 0.0 <= perf[0] <= 100.0
