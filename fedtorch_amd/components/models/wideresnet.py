@@ -1,7 +1,11 @@
 # -*- coding: utf-8 -*-
-"""WideResNet for CIFAR (parity with reference `nonconvex/wideresnet.py`)."""
+"""WideResNet for CIFAR (parity with reference `nonconvex/wideresnet.py`).
+Pre-activation relu(bn(x)) pairs use BNReLU so the fused NHWC BN kernels
+fold the ReLU in on GPU (ops/batchnorm.py)."""
 import torch.nn as nn
 import torch.nn.functional as F
+
+from fedtorch_amd.ops.batchnorm import BNReLU
 
 _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10}
 
@@ -9,10 +13,10 @@ _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10}
 class _WideBlock(nn.Module):
     def __init__(self, in_planes, out_planes, stride, drop_rate=0.0):
         super().__init__()
-        self.bn1 = nn.BatchNorm2d(in_planes)
+        self.bn1 = BNReLU(in_planes)
         self.conv1 = nn.Conv2d(in_planes, out_planes, kernel_size=3,
                                stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(out_planes)
+        self.bn2 = BNReLU(out_planes)
         self.conv2 = nn.Conv2d(out_planes, out_planes, kernel_size=3,
                                stride=1, padding=1, bias=False)
         self.drop_rate = drop_rate
@@ -21,9 +25,9 @@ class _WideBlock(nn.Module):
             in_planes, out_planes, kernel_size=1, stride=stride, bias=False)
 
     def forward(self, x):
-        pre = F.relu(self.bn1(x))
-        out = self.conv1(pre if not self.equal_io else pre)
-        out = F.relu(self.bn2(out))
+        pre = self.bn1(x)
+        out = self.conv1(pre)
+        out = self.bn2(out)
         if self.drop_rate > 0:
             out = F.dropout(out, p=self.drop_rate, training=self.training)
         out = self.conv2(out)
@@ -48,7 +52,7 @@ class WideResNet(nn.Module):
                                          stride if i == 0 else 1, drop_rate))
                 in_planes = w
         self.blocks = nn.Sequential(*blocks)
-        self.bn_final = nn.BatchNorm2d(widths[3])
+        self.bn_final = BNReLU(widths[3])
         self.fc = nn.Linear(widths[3], self.num_classes)
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
@@ -60,7 +64,7 @@ class WideResNet(nn.Module):
 
     def forward(self, x):
         out = self.blocks(self.conv1(x))
-        out = F.relu(self.bn_final(out))
+        out = self.bn_final(out)
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.fc(out)
 
