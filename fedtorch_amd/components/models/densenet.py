@@ -1,5 +1,6 @@
 # -*- coding: utf-8 -*-
-"""DenseNet for CIFAR (parity with reference `nonconvex/densenet.py`).
+"""DenseNet (BN+ReLU pairs as BNReLU: fused NHWC BN kernels fold the
+ReLU on GPU) for CIFAR (parity with reference `nonconvex/densenet.py`).
 
 Supports plain and BC mode (bottleneck + compression), growth rate and
 compression per the reference factory (`densenet.py:200-208`).
@@ -10,19 +11,21 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from fedtorch_amd.ops.batchnorm import BNReLU
+
 _NUM_CLASSES = {'cifar10': 10, 'cifar100': 100, 'svhn': 10}
 
 
 class _DenseLayer(nn.Module):
     def __init__(self, num_channels, growth_rate, drop_rate):
         super().__init__()
-        self.bn1 = nn.BatchNorm2d(num_channels)
+        self.bn1 = BNReLU(num_channels)
         self.conv1 = nn.Conv2d(num_channels, growth_rate, kernel_size=3,
                                padding=1, bias=False)
         self.drop_rate = drop_rate
 
     def forward(self, x):
-        out = self.conv1(F.relu(self.bn1(x)))
+        out = self.conv1(self.bn1(x))
         if self.drop_rate > 0:
             out = F.dropout(out, p=self.drop_rate, training=self.training)
         return torch.cat([x, out], 1)
@@ -32,16 +35,16 @@ class _BottleneckLayer(nn.Module):
     def __init__(self, num_channels, growth_rate, drop_rate):
         super().__init__()
         inter = 4 * growth_rate
-        self.bn1 = nn.BatchNorm2d(num_channels)
+        self.bn1 = BNReLU(num_channels)
         self.conv1 = nn.Conv2d(num_channels, inter, kernel_size=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(inter)
+        self.bn2 = BNReLU(inter)
         self.conv2 = nn.Conv2d(inter, growth_rate, kernel_size=3, padding=1,
                                bias=False)
         self.drop_rate = drop_rate
 
     def forward(self, x):
-        out = self.conv1(F.relu(self.bn1(x)))
-        out = self.conv2(F.relu(self.bn2(out)))
+        out = self.conv1(self.bn1(x))
+        out = self.conv2(self.bn2(out))
         if self.drop_rate > 0:
             out = F.dropout(out, p=self.drop_rate, training=self.training)
         return torch.cat([x, out], 1)
@@ -50,13 +53,13 @@ class _BottleneckLayer(nn.Module):
 class _Transition(nn.Module):
     def __init__(self, num_channels, num_out_channels, drop_rate):
         super().__init__()
-        self.bn1 = nn.BatchNorm2d(num_channels)
+        self.bn1 = BNReLU(num_channels)
         self.conv1 = nn.Conv2d(num_channels, num_out_channels, kernel_size=1,
                                bias=False)
         self.drop_rate = drop_rate
 
     def forward(self, x):
-        out = self.conv1(F.relu(self.bn1(x)))
+        out = self.conv1(self.bn1(x))
         if self.drop_rate > 0:
             out = F.dropout(out, p=self.drop_rate, training=self.training)
         return F.avg_pool2d(out, 2)
@@ -85,7 +88,7 @@ class DenseNet(nn.Module):
                 blocks.append(_Transition(num_channels, out_ch, drop_rate))
                 num_channels = out_ch
         self.blocks = nn.Sequential(*blocks)
-        self.bn_final = nn.BatchNorm2d(num_channels)
+        self.bn_final = BNReLU(num_channels)
         self.classifier = nn.Linear(num_channels, self.num_classes)
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
@@ -97,7 +100,7 @@ class DenseNet(nn.Module):
 
     def forward(self, x):
         out = self.blocks(self.conv1(x))
-        out = F.relu(self.bn_final(out))
+        out = self.bn_final(out)
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.classifier(out)
 
