@@ -94,6 +94,8 @@ def main(args):
         if v is not None:
             cmd += [k, str(v)]
     print('\nRunning:\n' + ' '.join(shlex.quote(c) for c in cmd))
+    if args.dry_run:
+        return 0
     env = dict(os.environ)
     env.setdefault('HSA_ENABLE_IPC_MODE_LEGACY', '0')
     return subprocess.call(cmd, env=env)
@@ -141,4 +143,6 @@ if __name__ == '__main__':
     parser.add_argument('-td', '--tmp_dir', default='/tmp', type=str)
     parser.add_argument('--bf16', action='store_true')
     parser.add_argument('--master_port', default=29500, type=int)
+    parser.add_argument('--dry_run', action='store_true',
+                        help='print the torchrun command and exit')
     sys.exit(main(parser.parse_args()))

@@ -190,3 +190,34 @@ def test_checkpoint_best_and_epoch_copies(tmp_path):
     state['current_epoch'] = 3
     save_to_checkpoint(state, False, d, 'checkpoint.pth.tar', save_all=True)
     assert os.path.exists(os.path.join(d, 'checkpoint_epoch_3.pth.tar'))
+
+
+def test_run_dist_command_mapping(capsys):
+    """run_dist maps the reference's short flags (`run_mpi.py:25-105`) onto
+    a torchrun command with the full parameter list."""
+    import shlex
+    from types import SimpleNamespace
+    import run_dist
+    ns = SimpleNamespace(
+        num_epochs_per_comm=1, num_clients=4, dataset='cifar10',
+        data_path='./data', batch_size=50, num_comms=3, lr_gamma=0.1,
+        lr_mu=1, lr_sync=1.0, weight_decay=1e-4, iid=False, local_steps=10,
+        on_cuda=False, federated=True, federated_type='fedgate',
+        federated_drfa=False, drfa_gamma=0.1,
+        federated_sync_type='local_step', online_client_rate=0.5,
+        num_class_per_client=2, synthetic_params=[0.0, 0.0], quantized=True,
+        compressed=False, compressed_ratio=1.0, unbalanced=False,
+        fed_personal=False, fed_personal_alpha=0.0, fed_adaptive_alpha=False,
+        sensitive_feature=9, quantized_bits=8, fedprox_mu=0.002, arch=None,
+        tmp_dir='/tmp', bf16=False, master_port=29911, dry_run=True)
+    assert run_dist.main(ns) == 0
+    out = capsys.readouterr().out
+    cmd = shlex.split(out.split('Running:\n', 1)[1].splitlines()[0])
+    assert cmd[cmd.index('--nproc-per-node') + 1] == '4'
+    joined = ' '.join(cmd)
+    assert '--federated_type fedgate' in joined
+    assert '--quantized True' in joined
+    assert '--online_client_rate 0.5' in joined
+    assert '--num_class_per_client 2' in joined
+    assert '--local_step 10' in joined
+    assert 'fedtorch_amd.main' in joined
