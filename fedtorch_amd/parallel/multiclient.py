@@ -67,6 +67,15 @@ class ClientPack(object):
         self.mom_init = [False] * self.C
         self.train_loaders = [None] * self.C
         self.partial = torch.zeros(n, device=dev)
+        # per-client BatchNorm running stats (the reference's centered mode
+        # gives every ClientCentered its own module and therefore its own
+        # BN buffers; the shared compute module must swap them per client)
+        self.bufs = None
+        if arena.buf_flat is not None:
+            self.bufs = torch.zeros((self.C, arena.buf_flat.numel()),
+                                    device=dev)
+            for c in range(self.C):
+                self.bufs[c].copy_(arena.buf_flat)
         for c in range(self.C):
             self.replicas[c].copy_(arena.flat)
 
@@ -91,6 +100,8 @@ class ClientPack(object):
         local_step_fn(loader) -> local_steps performed."""
         client = self.base
         client.arena.load_flat(server_flat)
+        if self.bufs is not None:
+            client.arena.buf_flat.copy_(self.bufs[j])
         if self.in_mom is not None:
             client.optimizer.bind_state(in_buf=self.in_mom[j],
                                         in_init=self.mom_init[j])
@@ -98,7 +109,27 @@ class ClientPack(object):
         if self.in_mom is not None:
             self.mom_init[j] = True
         self.replicas[j].copy_(client.arena.flat)
+        if self.bufs is not None:
+            self.bufs[j].copy_(client.arena.buf_flat)
         return steps
+
+    def mean_buffers(self, online_local):
+        """Average the ONLINE local clients' BN stats into the compute
+        module's buffer arena (pre-all-reduce partial of the global mean)
+        and refresh every local client's stats with it after sync."""
+        if self.bufs is None or not online_local:
+            return
+        mean = self.bufs[list(online_local)].mean(0)
+        self.base.arena.buf_flat.copy_(mean)
+
+    def adopt_buffers(self):
+        """After the world-level BN-stat all-reduce every client adopts
+        the global mean (mirrors all one-client ranks ending the round
+        with identical buffers)."""
+        if self.bufs is None:
+            return
+        self.bufs.copy_(self.base.arena.buf_flat.unsqueeze(0)
+                        .expand_as(self.bufs))
 
     def accumulate_partial(self, server_flat, weights):
         """partial = sum_j weights[j] * (server - replica_j) — ONE batched

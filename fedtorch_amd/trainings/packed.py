@@ -164,6 +164,16 @@ def train_and_validate_federated_packed(client, pack, validate=False):
         client.comm.all_reduce(partial)
         _apply_aggregate(client, server, partial)
         client.arena.load_flat(server)
+        # BN running stats: rank-local mean over online local clients, then
+        # the world-level average, then every client adopts the mean (same
+        # end state as W*C one-client ranks)
+        online_local = [j for j in range(pack.C) if weights[j] != 0.0]
+        pack.mean_buffers(online_local)
+        from fedtorch_amd.aggregation.federated import aggregate_bn_buffers
+        aggregate_bn_buffers(args, client.comm, client.arena,
+                             list(range(args.graph.n_nodes)),
+                             work=client.work)
+        pack.adopt_buffers()
         for j in range(pack.C):
             # every client re-syncs at round end (its next round reloads
             # the server anyway; keeps replicas bounded in staleness)

@@ -59,3 +59,48 @@ def test_run_client_swaps_state():
     assert pack.mom_init == [True, True]
     assert torch.allclose(pack.in_mom[0], pack.in_mom[1])
     assert pack.in_mom[0].abs().sum() > 0
+
+
+def test_pack_per_client_bn_buffers():
+    """Each virtual client keeps its OWN BatchNorm running stats across the
+    shared compute module (the reference's centered mode gives every
+    client its own module; `parallel/multiclient.py` swaps buffer rows)."""
+    import torch
+    import torch.nn as nn
+    from types import SimpleNamespace
+    from fedtorch_amd.parallel.arena import Arena
+    from fedtorch_amd.components.optim.sgd import FusedSGD
+
+    torch.manual_seed(11)
+    m = nn.Sequential(nn.Conv2d(3, 4, 3, padding=1), nn.BatchNorm2d(4))
+    arena = Arena(m)
+    opt = FusedSGD(arena, lr=0.1, in_momentum=0.9)
+    client = SimpleNamespace(
+        arena=arena, optimizer=opt, model=m,
+        args=SimpleNamespace(graph=SimpleNamespace(rank=0, n_nodes=2)))
+    from fedtorch_amd.parallel.multiclient import ClientPack
+    pack = ClientPack(client, 2)
+    assert pack.bufs is not None and pack.bufs.shape[0] == 2
+
+    server = arena.clone_flat()
+
+    def steps_with_data(x):
+        def run(loader):
+            m(x)  # train-mode forward updates running stats
+            return 1
+        return run
+
+    x0 = torch.randn(8, 3, 5, 5) * 3 + 7    # big mean/var
+    x1 = torch.randn(8, 3, 5, 5) * 0.1      # small
+    pack.train_loaders = [None, None]
+    pack.run_client(0, server, steps_with_data(x0))
+    pack.run_client(1, server, steps_with_data(x1))
+    assert not torch.allclose(pack.bufs[0], pack.bufs[1])
+    # client 0's stats reflect the large-mean batch, client 1's don't
+    assert pack.bufs[0].abs().max() > pack.bufs[1].abs().max()
+    # adopt_buffers: all rows equal the module's buffer arena
+    arena.buf_flat.fill_(0.5)
+    pack.adopt_buffers()
+    assert torch.allclose(pack.bufs[0], pack.bufs[1])
+    assert torch.allclose(pack.bufs[0],
+                          torch.full_like(pack.bufs[0], 0.5))
