@@ -83,3 +83,34 @@ def test_sync_freq_turn_off():
         lr_change_epochs='5')
     assert all(s == 4 for s in steps[:5])
     assert all(s == 1 for s in steps[5:])
+
+
+def test_parameter_derivations_match_reference():
+    """Post-parse derivations (reference `parameters.py:240-260`)."""
+    import pytest as _pytest
+    from fedtorch_amd.parameters import get_args
+    base = ['-d', 'mnist', '-a', 'mlp', '-f', 'true', '--num_comms', '10',
+            '--num_epochs_per_comm', '2', '--online_client_rate', '0.5',
+            '--checkpoint', '/tmp/ft_deriv', '--debug', 'false']
+    a = get_args(base + ['--federated_type', 'fedavg'])
+    # num_epochs = per_comm * comms * online_rate (`:248`)
+    assert a.num_epochs == int(2 * 10 * 0.5)
+    # afl forces local_step sync with step 1 (`:250-252`)
+    a = get_args(base + ['--federated_type', 'afl'])
+    assert a.federated_sync_type == 'local_step' and a.local_step == 1
+    # personalization forces fed_personal (`:257-259`)
+    for t in ('apfl', 'perfedme', 'perfedavg'):
+        a = get_args(base + ['--federated_type', t])
+        assert a.fed_personal, t
+    # qsparse implies compression (the reference's `:253` is a no-op bug —
+    # fixed here, documented in parameters.py)
+    a = get_args(base + ['--federated_type', 'qsparse'])
+    assert a.compressed
+    # quantize + compress are mutually exclusive (`:254-255`)
+    with _pytest.raises(ValueError):
+        get_args(base + ['--federated_type', 'fedavg', '--quantized', 'true',
+                         '--compressed', 'true'])
+    # reshuffle is rejected under federation (`:246-247`)
+    with _pytest.raises(ValueError):
+        get_args(base + ['--federated_type', 'fedavg',
+                         '--reshuffle_per_epoch', 'true'])
