@@ -55,18 +55,28 @@ def _buf(work, name, like):
     return work[name]
 
 
-def aggregate_bn_buffers(args, comm, arena, online_clients, work=None):
+def aggregate_bn_buffers(args, comm, arena, online_clients, work=None,
+                         prescaled=False):
     """Average BatchNorm running stats over the ONLINE clients (uniform
     mean; offline ranks contribute zero and adopt the mean).  The reference
-    never syncs buffers — see `parallel/arena.py` docstring."""
+    never syncs buffers — see `parallel/arena.py` docstring.
+
+    ``prescaled=True`` (packed mode): the buffer arena already holds this
+    rank's weighted partial (sum over its online clients / total_online,
+    `ClientPack.partial_buffers`), so the all-reduce runs without further
+    scaling — ranks with unequal online-client counts contribute with the
+    correct proportional weight."""
     if arena.buf_flat is None or \
             not getattr(args, 'aggregate_bn_stats', True):
         return
     work = work if work is not None else {}
-    w = 1.0 / len(online_clients) \
-        if args.graph.rank in online_clients else 0.0
     tmp = _buf(work, 'bn_stats_buf', arena.buf_flat)
-    tmp.copy_(arena.buf_flat).mul_(w)
+    if prescaled:
+        tmp.copy_(arena.buf_flat)
+    else:
+        w = 1.0 / len(online_clients) \
+            if args.graph.rank in online_clients else 0.0
+        tmp.copy_(arena.buf_flat).mul_(w)
     comm.all_reduce(tmp)
     arena.buf_flat.copy_(tmp)
 

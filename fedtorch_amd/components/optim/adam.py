@@ -2,7 +2,17 @@
 """AdamW over a flat arena (parity with reference
 `components/optimizers/adam.py:48-104`, including its dual-use convention:
 ``apply_lr=False`` applies ``p -= scale * grad`` BEFORE the moment updates,
-which is how the reference uses it at sync)."""
+which is how the reference uses it at sync).
+
+Documented divergence: the reference (`adam.py:97-99`) applies the
+moment-based ``addcdiv`` step UNCONDITIONALLY, i.e. even on the sync call
+(`apply_lr=False`) it takes a second, lr-sized moment step on top of
+``p -= scale*grad``.  Here the moment step is gated on ``apply_lr`` — the
+sync call only applies the scaled aggregate (and still refreshes the
+moments with it), which is the behavior every other optimizer in both
+codebases has at sync.  This changes adam/fedadam sync trajectories
+relative to the reference by design (like the other deliberate bug fixes
+listed in `aggregation/federated.py:14-25`)."""
 import math
 
 import torch

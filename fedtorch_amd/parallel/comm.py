@@ -79,15 +79,25 @@ class Comm(object):
 
     def all_gather_flat(self, flat):
         """All-gather equal-size flat tensors -> [W, N] stacked tensor.
-        (gloo requires a flat output buffer, hence the view.)"""
+        (gloo requires a flat output buffer, hence the view.)
+
+        int16 payloads (16-bit quantization, `ops.quantize(num_bits=16)`)
+        are gathered as their int8 byte view: RCCL/ProcessGroupNCCL has no
+        int16 dtype mapping, and a gather is bytes-only anyway."""
         if not dist_ready():
             return flat.unsqueeze(0)
+        shape = tuple(flat.shape)
+        as_i16 = flat.dtype == torch.int16
+        if as_i16:
+            flat = flat.contiguous().view(torch.int8)
         world = dist.get_world_size()
         out = flat.new_empty(world * flat.numel())
         tok = self._tic()
         dist.all_gather_into_tensor(out, flat)
         self._toc(tok)
-        return out.view((world,) + tuple(flat.shape))
+        if as_i16:
+            out = out.view(torch.int16)
+        return out.view((world,) + shape)
 
     def barrier(self):
         if dist_ready():

@@ -113,14 +113,21 @@ class ClientPack(object):
             self.bufs[j].copy_(client.arena.buf_flat)
         return steps
 
-    def mean_buffers(self, online_local):
-        """Average the ONLINE local clients' BN stats into the compute
-        module's buffer arena (pre-all-reduce partial of the global mean)
-        and refresh every local client's stats with it after sync."""
-        if self.bufs is None or not online_local:
+    def partial_buffers(self, online_local, total_online):
+        """Write this rank's PRE-SCALED partial of the global BN-stat mean
+        (sum over ONLINE local clients / total_online) into the compute
+        module's buffer arena.  Zeroed when no local client is online, so
+        a plain world all-reduce of the buffer arena yields the mean over
+        ALL online clients regardless of how they spread across ranks
+        (ranks with more online clients weigh proportionally more)."""
+        if self.bufs is None:
             return
-        mean = self.bufs[list(online_local)].mean(0)
-        self.base.arena.buf_flat.copy_(mean)
+        buf = self.base.arena.buf_flat
+        if not online_local or total_online <= 0:
+            buf.zero_()
+            return
+        torch.sum(self.bufs[list(online_local)], dim=0, out=buf)
+        buf.div_(float(total_online))
 
     def adopt_buffers(self):
         """After the world-level BN-stat all-reduce every client adopts

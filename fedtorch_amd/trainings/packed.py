@@ -164,15 +164,18 @@ def train_and_validate_federated_packed(client, pack, validate=False):
         client.comm.all_reduce(partial)
         _apply_aggregate(client, server, partial)
         client.arena.load_flat(server)
-        # BN running stats: rank-local mean over online local clients, then
-        # the world-level average, then every client adopts the mean (same
-        # end state as W*C one-client ranks)
-        online_local = [j for j in range(pack.C) if weights[j] != 0.0]
-        pack.mean_buffers(online_local)
+        # BN running stats: each rank contributes its pre-scaled partial
+        # (sum over its online clients / total online) so the plain world
+        # all-reduce gives the mean over ALL online clients even when
+        # ranks have unequal online counts; then every client adopts the
+        # mean (same end state as W*C one-client ranks)
+        online_local = [j for j in range(pack.C)
+                        if pack.global_id(j) in online]
+        pack.partial_buffers(online_local, len(online))
         from fedtorch_amd.aggregation.federated import aggregate_bn_buffers
         aggregate_bn_buffers(args, client.comm, client.arena,
                              list(range(args.graph.n_nodes)),
-                             work=client.work)
+                             work=client.work, prescaled=True)
         pack.adopt_buffers()
         for j in range(pack.C):
             # every client re-syncs at round end (its next round reloads

@@ -104,3 +104,43 @@ def test_pack_per_client_bn_buffers():
     assert torch.allclose(pack.bufs[0], pack.bufs[1])
     assert torch.allclose(pack.bufs[0],
                           torch.full_like(pack.bufs[0], 0.5))
+
+
+def test_partial_buffers_weighted_by_online_count():
+    """the BN partial must be weighted by each rank's
+    online-client count, not a uniform per-rank mean.  Two packs with 2
+    and 1 online clients (3 online total): the SUM of the two prescaled
+    partials must equal the mean over the 3 online clients' stats, and a
+    rank with zero online clients must contribute exact zeros (not its
+    stale compute-module buffer)."""
+    import torch.nn as nn
+    from types import SimpleNamespace
+
+    def make_pack(rank, C):
+        torch.manual_seed(100 + rank)
+        m = nn.Sequential(nn.Conv2d(3, 4, 3, padding=1), nn.BatchNorm2d(4))
+        arena = Arena(m)
+        opt = FusedSGD(arena, lr=0.1, in_momentum=0.9)
+        client = SimpleNamespace(
+            arena=arena, optimizer=opt, model=m,
+            args=SimpleNamespace(graph=SimpleNamespace(rank=rank,
+                                                       n_nodes=2)))
+        pack = ClientPack(client, C)
+        for c in range(C):
+            pack.bufs[c].normal_(float(rank * 10 + c), 1.0)
+        return pack
+
+    SimpleNamespace  # noqa: B018
+    p0, p1 = make_pack(0, 2), make_pack(1, 2)
+    total_online = 3
+    p0.partial_buffers([0, 1], total_online)      # both local online
+    p1.partial_buffers([1], total_online)         # one local online
+    got = p0.base.arena.buf_flat + p1.base.arena.buf_flat
+    want = (p0.bufs[0] + p0.bufs[1] + p1.bufs[1]) / 3.0
+    assert torch.allclose(got, want, atol=1e-6)
+
+    # zero online local clients -> exact zero contribution
+    p1.base.arena.buf_flat.fill_(123.0)  # stale garbage
+    p1.partial_buffers([], total_online)
+    assert torch.equal(p1.base.arena.buf_flat,
+                       torch.zeros_like(p1.base.arena.buf_flat))
