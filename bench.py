@@ -275,13 +275,15 @@ def main():
     # driver bench measured 3.1 ms/step vs 1.64 steady for exactly this
     # reason).  So: run full rounds here until per-round time stabilizes
     # (<=8 rounds), then do the W contractual warmup steps.
-    if on_gpu:
+    if on_gpu or os.environ.get('FEDTORCH_FORCE_SETTLE') == '1':
         prev = None
         for r_i in range(8):
-            torch.cuda.synchronize()
+            if on_gpu:
+                torch.cuda.synchronize()
             ts = time.perf_counter()
             run(TAU)
-            torch.cuda.synchronize()
+            if on_gpu:
+                torch.cuda.synchronize()
             dt = time.perf_counter() - ts
             if dist.is_initialized():
                 # all ranks must take the SAME number of settle rounds

@@ -221,3 +221,32 @@ def test_run_dist_command_mapping(capsys):
     assert '--num_class_per_client 2' in joined
     assert '--local_step 10' in joined
     assert 'fedtorch_amd.main' in joined
+
+
+def test_bench_four_rank_cpu_contract_with_settle():
+    """8-GPU readiness: the driver may launch
+    `torchrun --nproc-per-node 8 bench.py --gpus 8`.  Exercise the
+    multi-rank path at world 4 on gloo INCLUDING the collective settle
+    loop (all ranks must agree on the number of settle rounds or the
+    sync collectives deadlock)."""
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR='127.0.0.1', MASTER_PORT='29885',
+               FEDTORCH_FORCE_SETTLE='1')
+    out = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '4', '--master-addr', '127.0.0.1',
+         '--master-port', '29886', os.path.join(repo, 'bench.py'),
+         '--gpus', '4', '--steps', '10', '--warmup', '1', '--batch', '8'],
+        capture_output=True, text=True, timeout=900, cwd=repo, env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith('{')]
+    assert len(lines) == 1, lines
+    d = json.loads(lines[0])
+    assert d['n_gpus'] == 4
+    assert d['config']['global_batch'] == 32
+    assert d['config']['parallelism'] == 'fedavg_dp4_tau10'
+    assert d['value'] > 0
