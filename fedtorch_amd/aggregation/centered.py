@@ -99,6 +99,9 @@ def scaffold_aggregation_centered(Clients, Server, online_clients,
     agg.zero_()
     diff = Server.work.setdefault('diff', Server.arena.new_buffer())
     ctrl_new = Server.work.setdefault('ctrl_new', Server.arena.new_buffer())
+    ctrl_delta = Server.work.setdefault('ctrl_delta',
+                                        Server.arena.new_buffer())
+    ctrl_delta.zero_()
     for o in online_clients:
         wo = (1.0 / n_online) if lambda_weight is None else \
             float(lambda_weight[o]) * args.graph.n_nodes / n_online
@@ -109,13 +112,17 @@ def scaffold_aggregation_centered(Clients, Server, online_clients,
                                     1.0 / (local_steps * lr))
         ops.scaled_diff(Server.arena.flat, Clients[o].arena.flat, diff, wo)
         agg.add_(diff)
-        # server control += (c+ - c) / num_workers
-        ops.axpby(Server.model_server_control, ctrl_new,
-                  a=1.0 / args.num_workers, b=1.0)
-        ops.axpby(Server.model_server_control,
-                  Clients[o].model_client_control,
-                  a=-1.0 / args.num_workers, b=1.0)
+        # accumulate (c+ - c); applied to the server control AFTER the loop
+        # so every client's c+ uses the round-start server control — the
+        # reference centered `scaffold.py:33` updates scp mid-loop, which
+        # makes later clients see earlier clients' deltas and its own two
+        # execution modes disagree; the distributed (parallel) semantics
+        # is the real federated behavior (oracle: tests/test_equivalence.py)
+        ctrl_delta.add_(ctrl_new).sub_(Clients[o].model_client_control)
         Clients[o].model_client_control.copy_(ctrl_new)
+    # server control += sum (c+ - c) / num_workers
+    ops.axpby(Server.model_server_control, ctrl_delta,
+              a=1.0 / args.num_workers, b=1.0)
     Server.optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
                           apply_in_momentum=False,
                           apply_out_momentum=args.out_momentum, grad=agg)

@@ -223,9 +223,14 @@ def scaffold_aggregation(args, comm, arena, server_flat, server_ctrl,
         b.zero_()
     ops.weighted_diff_restore(server_flat, arena.flat, a, w)
     comm.all_reduce(pair)
-    # server control: -= sum_ctrl_delta * |S|/N (`scaffold.py:66`, fixed: N =
-    # n_nodes, and the GATHERED aggregate is used).
-    server_ctrl.sub_(b, alpha=len(online_clients) / args.graph.n_nodes)
+    # server control += sum_i w_i (c_i+ - c_i) * |S|/N  — the reference
+    # carries (c - c+)*w and does `scp -= d[1]*|S|/N` (`scaffold.py:36,61`),
+    # i.e. a net PLUS of the control delta (SCAFFOLD paper eq. 5); here b
+    # holds +(c+ - c)*w so this adds (fixed vs r1: the sign was flipped,
+    # caught by the centered==distributed oracle).  Also fixed vs the
+    # reference: N = n_nodes and the GATHERED aggregate is used
+    # (`scaffold.py:59-64` overwrites it with the local tensor).
+    server_ctrl.add_(b, alpha=len(online_clients) / args.graph.n_nodes)
     if ctrl_new is not None:
         client_ctrl.copy_(ctrl_new)
     optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,

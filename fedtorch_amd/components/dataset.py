@@ -85,9 +85,19 @@ def partitioner(args, dataset, shuffle, world_size, partition_type='normal',
     return part.use(args.graph.rank)
 
 
-def _make_loader(args, data, batch_size, shuffle, drop_last=False):
+def _make_loader(args, data, batch_size, shuffle, drop_last=False, tag=0):
+    # Shuffle order is a pure function of (manual_seed, client id, tag):
+    # batch order on distributed rank i matches centered/packed virtual
+    # client i exactly — the centered==distributed equivalence oracle
+    #  depends on this, and it decouples loader order from
+    # unrelated global-RNG consumption.
+    gen = None
+    if shuffle:
+        gen = torch.Generator()
+        gen.manual_seed(int(args.manual_seed) * 1000003 +
+                        int(args.graph.rank) * 8191 + tag)
     return torch.utils.data.DataLoader(
-        data, batch_size=batch_size, shuffle=shuffle,
+        data, batch_size=batch_size, shuffle=shuffle, generator=gen,
         num_workers=args.num_workers, pin_memory=args.pin_memory,
         drop_last=drop_last,
         persistent_workers=args.num_workers > 0)
@@ -199,11 +209,12 @@ def partition_dataset(args, shuffle, dataset_type, Partitioner=None,
                             [len(data_to_load) - val_size, val_size])
             data_loader = [
                 _make_loader(args, data_to_load_train, batch_size, True),
-                _make_loader(args, data_to_load_val, batch_size, True)]
+                _make_loader(args, data_to_load_val, batch_size, True, tag=1)]
             if args.federated_type == 'perfedavg':
                 data_loader = [
                     data_loader[0],
-                    _make_loader(args, data_to_load_val1, batch_size, True),
+                    _make_loader(args, data_to_load_val1, batch_size, True,
+                                 tag=2),
                     data_loader[1]]
                 # reference order is [train, val, val1]; keep (train, val,
                 # val1) via define_dataset unpacking:

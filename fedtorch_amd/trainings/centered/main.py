@@ -102,7 +102,7 @@ def run_client_local_steps(client, Server, tracker, val_batch=None):
     return local_steps, lr
 
 
-def train_and_validate_federated_centered(Clients, Server):
+def train_and_validate_federated_centered(Clients, Server, validate=True):
     log('start training and validation with Federated setting in a '
         'centered way.')
     tracker = define_local_training_tracker()
@@ -145,11 +145,13 @@ def train_and_validate_federated_centered(Clients, Server):
                     Clients[oc].full_loss += loss.item()
 
             # pre-training validation of the server model on client data
-            do_validate_centered(
-                Clients[oc].args, Server.model, Server.criterion,
-                Server.metrics, Server.optimizer, Clients[oc].train_loader,
-                Server.global_val_tracker, val=False, local=False)
-            if args.per_class_acc:
+            if validate:
+                do_validate_centered(
+                    Clients[oc].args, Server.model, Server.criterion,
+                    Server.metrics, Server.optimizer,
+                    Clients[oc].train_loader,
+                    Server.global_val_tracker, val=False, local=False)
+            if validate and args.per_class_acc:
                 # per-client trackers feed the worst/best/var lines
                 # (reference `centered/main.py:77-88`)
                 Clients[oc].reset_tracker(Clients[oc].local_val_tracker)
@@ -159,7 +161,7 @@ def train_and_validate_federated_centered(Clients, Server):
                     Server.metrics, Server.optimizer,
                     Clients[oc].train_loader,
                     Clients[oc].global_val_tracker, val=False, local=False)
-            if args.fed_personal:
+            if validate and args.fed_personal:
                 do_validate_centered(
                     Clients[oc].args, Server.model, Server.criterion,
                     Server.metrics, Server.optimizer,
@@ -179,18 +181,20 @@ def train_and_validate_federated_centered(Clients, Server):
                 Clients[oc], Server, tracker, val_batch=val_batch)
 
             # post-training validation of the client model
-            do_validate_centered(
-                Clients[oc].args, Clients[oc].model, Clients[oc].criterion,
-                Clients[oc].metrics, Clients[oc].optimizer,
-                Clients[oc].train_loader, Server.local_val_tracker,
-                val=False, local=True)
-            if args.per_class_acc:
+            if validate:
+                do_validate_centered(
+                    Clients[oc].args, Clients[oc].model,
+                    Clients[oc].criterion, Clients[oc].metrics,
+                    Clients[oc].optimizer,
+                    Clients[oc].train_loader, Server.local_val_tracker,
+                    val=False, local=True)
+            if validate and args.per_class_acc:
                 do_validate_centered(
                     Clients[oc].args, Clients[oc].model,
                     Clients[oc].criterion, Clients[oc].metrics,
                     Clients[oc].optimizer, Clients[oc].train_loader,
                     Clients[oc].local_val_tracker, val=False, local=True)
-            if args.fed_personal:
+            if validate and args.fed_personal:
                 do_validate_centered(
                     Clients[oc].args, Clients[oc].model,
                     Clients[oc].criterion, Clients[oc].metrics,
@@ -233,10 +237,12 @@ def train_and_validate_federated_centered(Clients, Server):
             log_validation_per_client_centered(args, Clients, online_clients,
                                                val=False, local=True)
 
-        do_validate_centered(args, Server.model, Server.criterion,
-                             Server.metrics, Server.optimizer,
-                             Server.test_loader, Server.global_test_tracker,
-                             val=False, local=False)
-        log_test_centered(args, Server.global_test_tracker)
+        if validate:
+            do_validate_centered(args, Server.model, Server.criterion,
+                                 Server.metrics, Server.optimizer,
+                                 Server.test_loader,
+                                 Server.global_test_tracker,
+                                 val=False, local=False)
+            log_test_centered(args, Server.global_test_tracker)
         logging_globally(tracker, start_global_time)
         start_global_time = time.time()
