@@ -11,8 +11,16 @@ def get_dataset(args, name, datasets_path, split='train'):
     if name in ('cifar10', 'cifar100', 'mnist', 'fashion_mnist', 'stl10'):
         return sources.get_vision_dataset(name, datasets_path, split)
     if name in ('emnist', 'emnist_full'):
-        # federated EMNIST: per-client shards; synthetic fallback shards by
-        # vision generator seeded per client.
+        # federated EMNIST: real per-writer shards when materialized
+        # (reference layout, `loader/federated_datasets.py:83-138`;
+        # see `federated_shards.py`), else deterministic synthetic
+        # stand-in shards seeded per client.
+        import os
+        from fedtorch_amd.components.datasets import federated_shards as fs
+        root = os.path.join(datasets_path, name)
+        if fs.emnist_shards_present(root):
+            return fs.EMNISTShards(root, split=split,
+                                   client_id=args.graph.rank)
         return sources.get_vision_dataset(
             name, datasets_path, split,
             seed=1234 + (args.graph.rank if split == 'train' else -1))

@@ -268,6 +268,12 @@ class CharWindows(torch.utils.data.Dataset):
 
 def get_shakespeare_dataset(args, split, client_id):
     root = os.path.join(args.data_dir, 'shakespeare')
+    # real per-role shards (reference layout,
+    # `loader/federated_datasets.py:390-472`) take priority
+    from fedtorch_amd.components.datasets import federated_shards as fs
+    if fs.shakespeare_shards_present(root):
+        return fs.ShakespeareShards(root, split=split, client_id=client_id,
+                                    seq_len=args.rnn_seq_len)
     pt = os.path.join(root, 'Client_{}.pt'.format(client_id)
                       if split == 'train' else 'Test.pt')
     if os.path.exists(pt):
