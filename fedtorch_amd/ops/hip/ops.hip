@@ -15,6 +15,7 @@
 #include "batchnorm.h"
 #include "stemconv.h"
 #include "convwrw.h"
+#include "convfwd.h"
 
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
 #define STREAM at::hip::getCurrentHIPStream().stream()
@@ -862,6 +863,68 @@ void cast_to_half(torch::Tensor src, torch::Tensor dst) {
 }
 
 // ==========================================================================
+// MFMA direct NHWC 3x3/s1/p1 conv FORWARD with fused BN prologue/epilogue
+// (convfwd.h)
+// ==========================================================================
+torch::Tensor conv3x3_bn_fwd(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor ysum, torch::Tensor in_a,
+                             torch::Tensor in_b, torch::Tensor res,
+                             bool relu_in) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 4,
+              "conv3x3_bn_fwd: 4D GPU tensors");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3_bn_fwd: channels_last only");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+                  w.scalar_type() == torch::kBFloat16,
+              "conv3x3_bn_fwd: bf16 only");
+  const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = w.size(0);
+  const bool ok = (Co == Ci) && ((Co == 16 && W == 32) ||
+                                 (Co == 32 && W == 16) ||
+                                 (Co == 64 && W == 8));
+  TORCH_CHECK(ok, "conv3x3_bn_fwd: unsupported (Co,Ci,W)=",
+              Co, ",", Ci, ",", W);
+  TORCH_CHECK(H % 8 == 0, "conv3x3_bn_fwd: H % 8 != 0");
+  auto y = torch::empty(
+      {N, Co, H, W},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  float* ysp = ysum.defined() && ysum.numel() > 0
+                   ? ysum.data_ptr<float>() : nullptr;
+  const float* ap = in_a.defined() && in_a.numel() > 0
+                        ? in_a.data_ptr<float>() : nullptr;
+  const float* bp = in_b.defined() && in_b.numel() > 0
+                        ? in_b.data_ptr<float>() : nullptr;
+  const __hip_bfloat16* rp =
+      res.defined() && res.numel() > 0
+          ? reinterpret_cast<const __hip_bfloat16*>(res.data_ptr())
+          : nullptr;
+  TORCH_CHECK((ap == nullptr) == (bp == nullptr),
+              "conv3x3_bn_fwd: in_a and in_b go together");
+  const __hip_bfloat16* xp =
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());
+  const __hip_bfloat16* wp =
+      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
+  __hip_bfloat16* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
+  if (Co == 16) {
+    const int grid = N * (H / 8) * 1;
+    hipLaunchKernelGGL((conv3x3_bn_fwd_k<16, 16, 16, 32>), dim3(grid),
+                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
+                       rp, N, H, relu_in ? 1 : 0);
+  } else if (Co == 32) {
+    const int grid = N * (H / 8) * 1;
+    hipLaunchKernelGGL((conv3x3_bn_fwd_k<32, 32, 32, 16>), dim3(grid),
+                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
+                       rp, N, H, relu_in ? 1 : 0);
+  } else {
+    const int grid = N * (H / 8) * 2;
+    hipLaunchKernelGGL((conv3x3_bn_fwd_k<64, 64, 32, 8>), dim3(grid),
+                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
+                       rp, N, H, relu_in ? 1 : 0);
+  }
+  return y;
+}
+
+// ==========================================================================
 // MFMA 3x3/s1/p1 NHWC bf16 conv weight gradient (convwrw.h)
 // ==========================================================================
 torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
@@ -1310,4 +1373,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cast_to_half", &cast_to_half);
   m.def("mfma_probe", &mfma_probe);
   m.def("conv3x3_wrw", &conv3x3_wrw);
+  m.def("conv3x3_bn_fwd", &conv3x3_bn_fwd);
 }

