@@ -31,7 +31,7 @@
 
 // bf16x8 / f32x4 come from convwrw.h (same TU)
 
-template <int CI, int CO, int CO_TILE, int W>
+template <int CI, int CO, int CO_TILE, int W, bool FUSE_IN, bool HAS_RES>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
     const __hip_bfloat16* __restrict__ x,   // [N, H, W, CI]
     const __hip_bfloat16* __restrict__ w,   // [CO, 3, 3, CI]
@@ -56,6 +56,7 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
 
   __shared__ __hip_bfloat16 xs[NSLOT * CIP];
   __shared__ __hip_bfloat16 bw[K8 * CO_TILE * 8];
+  __shared__ float sa[CI], sb[CI];
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE, lane = tid & (WAVE - 1);
@@ -67,6 +68,11 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
   const int n = blockIdx.x / ((CO / CO_TILE) * (H / R));
   const int co0 = cob * CO_TILE;
   const int h0 = rb * R;
+
+  if (FUSE_IN && tid < CI) {
+    sa[tid] = in_a[tid];
+    sb[tid] = in_b[tid];
+  }
 
   // ---- stage weights: bw[k/8][co][j] = w[co0+co][tap][ci], k=tap*CI+ci --
   for (int e = tid; e < K8 * CO_TILE; e += FT_BLOCK) {
@@ -112,23 +118,32 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
           x + (((long)n * H + hh) * W) * CI + c * 8);
     const int pix = c * 8 / CI;          // pixel within the row
     const int ci0 = c * 8 % CI;
-    if (in_a != nullptr) {
-      // x~ = [relu](a*x + b [+ res]) fused into the staging store
-      uint4 rv = {0, 0, 0, 0};
-      if (res != nullptr && hh >= 0 && hh < H)
-        rv = *reinterpret_cast<const uint4*>(
-            res + (((long)n * H + hh) * W) * CI + c * 8);
-      const __hip_bfloat16* xv = reinterpret_cast<const __hip_bfloat16*>(&v);
-      const __hip_bfloat16* rr = reinterpret_cast<const __hip_bfloat16*>(&rv);
-      __hip_bfloat16 out[8];
+    if (FUSE_IN) {
+      // x~ = [relu](a*x + b [+ res]) fused into the staging store.
+      // Out-of-bounds rows stay ZERO: conv zero-pads the TRANSFORMED
+      // input, so the transform must not touch the halo.  FUSE_IN and
+      // HAS_RES are COMPILE-TIME: a runtime branch here makes hipcc
+      // branch around each staging load and drain vmcnt per element
+      // (guide §5 trap (c); measured +12 us per call).
+      if (hh >= 0 && hh < H) {
+        uint4 rv = {0, 0, 0, 0};
+        if (HAS_RES)
+          rv = *reinterpret_cast<const uint4*>(
+              res + (((long)n * H + hh) * W) * CI + c * 8);
+        const __hip_bfloat16* xv =
+            reinterpret_cast<const __hip_bfloat16*>(&v);
+        const __hip_bfloat16* rr =
+            reinterpret_cast<const __hip_bfloat16*>(&rv);
+        __hip_bfloat16 out[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float t = fmaf(in_a[ci0 + j], (float)xv[j], in_b[ci0 + j]);
-        if (res != nullptr) t += (float)rr[j];
-        if (relu_in) t = fmaxf(t, zero);
-        out[j] = (__hip_bfloat16)t;
+        for (int j = 0; j < 8; ++j) {
+          float t = fmaf(sa[ci0 + j], (float)xv[j], sb[ci0 + j]);
+          if (HAS_RES) t += (float)rr[j];
+          if (relu_in) t = fmaxf(t, zero);
+          out[j] = (__hip_bfloat16)t;
+        }
+        v = *reinterpret_cast<const uint4*>(out);
       }
-      v = *reinterpret_cast<const uint4*>(out);
     }
     *reinterpret_cast<uint4*>(
         &xs[(long)(row * XC + 1 + pix) * CIP + ci0]) = v;
@@ -188,6 +203,9 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
     }
   }
   if (ysum != nullptr) {
+    // per-WG partial row in bnh_norm_k's [B, C, 2] layout (B = gridDim.x):
+    // NO atomics — 1024 WGs atomically adding to 2*CO words measured
+    // +12 us/call; bnh_norm_k's block-parallel finalize eats the rows.
     // reduce lanes that share fm (kg = 0..3): xor over bits 4,5 of lane
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -206,16 +224,19 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
       }
     }
     __syncthreads();
-    if (tid < NT * 16) {
-      const int nt = tid / 16, ch = tid % 16;
+    float* prow = ysum + (long)blockIdx.x * CO * 2;
+    for (int c0 = tid; c0 < CO; c0 += FT_BLOCK) {
       float ts = 0.f, tss = 0.f;
+      if (c0 >= co0 && c0 < co0 + CO_TILE) {
+        const int nt = (c0 - co0) / 16, ch = (c0 - co0) % 16;
 #pragma unroll
-      for (int wv = 0; wv < 4; ++wv) {
-        ts += red[wv][nt][ch][0];
-        tss += red[wv][nt][ch][1];
+        for (int wv = 0; wv < 4; ++wv) {
+          ts += red[wv][nt][ch][0];
+          tss += red[wv][nt][ch][1];
+        }
       }
-      atomicAdd(&ysum[co0 + nt * 16 + ch], ts);
-      atomicAdd(&ysum[CO + co0 + nt * 16 + ch], tss);
+      prow[c0 * 2] = ts;
+      prow[c0 * 2 + 1] = tss;
     }
   }
 }

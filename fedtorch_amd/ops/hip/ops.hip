@@ -890,6 +890,13 @@ torch::Tensor conv3x3_bn_fwd(torch::Tensor x, torch::Tensor w,
       x.options().memory_format(at::MemoryFormat::ChannelsLast));
   float* ysp = ysum.defined() && ysum.numel() > 0
                    ? ysum.data_ptr<float>() : nullptr;
+  if (ysp) {
+    const long grid_total =
+        (long)N * (H / 8) * (Co == 64 ? 2 : 1);
+    TORCH_CHECK(ysum.numel() == grid_total * Co * 2,
+                "conv3x3_bn_fwd: ysum must be [grid, Co, 2] fp32, grid=",
+                grid_total);
+  }
   const float* ap = in_a.defined() && in_a.numel() > 0
                         ? in_a.data_ptr<float>() : nullptr;
   const float* bp = in_b.defined() && in_b.numel() > 0
@@ -905,22 +912,36 @@ torch::Tensor conv3x3_bn_fwd(torch::Tensor x, torch::Tensor w,
   const __hip_bfloat16* wp =
       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
   __hip_bfloat16* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
-  if (Co == 16) {
-    const int grid = N * (H / 8) * 1;
-    hipLaunchKernelGGL((conv3x3_bn_fwd_k<16, 16, 16, 32>), dim3(grid),
-                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
-                       rp, N, H, relu_in ? 1 : 0);
-  } else if (Co == 32) {
-    const int grid = N * (H / 8) * 1;
-    hipLaunchKernelGGL((conv3x3_bn_fwd_k<32, 32, 32, 16>), dim3(grid),
-                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
-                       rp, N, H, relu_in ? 1 : 0);
-  } else {
-    const int grid = N * (H / 8) * 2;
-    hipLaunchKernelGGL((conv3x3_bn_fwd_k<64, 64, 32, 8>), dim3(grid),
-                       dim3(FT_BLOCK), 0, STREAM, xp, wp, yp, ysp, ap, bp,
-                       rp, N, H, relu_in ? 1 : 0);
+  const bool fuse = ap != nullptr;
+  const bool hres = rp != nullptr;
+  TORCH_CHECK(!hres || fuse, "conv3x3_bn_fwd: residual needs in_a/in_b");
+#define FT_CONV_LAUNCH(CI_, CO_, COT_, W_, XN_)                            \
+  {                                                                        \
+    const int grid = N * (H / 8) * XN_;                                    \
+    if (fuse && hres)                                                      \
+      hipLaunchKernelGGL((conv3x3_bn_fwd_k<CI_, CO_, COT_, W_, true,       \
+                                           true>),                        \
+                         dim3(grid), dim3(FT_BLOCK), 0, STREAM, xp, wp,    \
+                         yp, ysp, ap, bp, rp, N, H, relu_in ? 1 : 0);      \
+    else if (fuse)                                                         \
+      hipLaunchKernelGGL((conv3x3_bn_fwd_k<CI_, CO_, COT_, W_, true,       \
+                                           false>),                       \
+                         dim3(grid), dim3(FT_BLOCK), 0, STREAM, xp, wp,    \
+                         yp, ysp, ap, bp, rp, N, H, relu_in ? 1 : 0);      \
+    else                                                                   \
+      hipLaunchKernelGGL((conv3x3_bn_fwd_k<CI_, CO_, COT_, W_, false,      \
+                                           false>),                       \
+                         dim3(grid), dim3(FT_BLOCK), 0, STREAM, xp, wp,    \
+                         yp, ysp, ap, bp, rp, N, H, relu_in ? 1 : 0);      \
   }
+  if (Co == 16) {
+    FT_CONV_LAUNCH(16, 16, 16, 32, 1)
+  } else if (Co == 32) {
+    FT_CONV_LAUNCH(32, 32, 32, 16, 1)
+  } else {
+    FT_CONV_LAUNCH(64, 64, 32, 8, 2)
+  }
+#undef FT_CONV_LAUNCH
   return y;
 }
 
