@@ -40,15 +40,26 @@ class _FusedBNFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum,
-                eps, relu, res):
+                eps, relu, res, part=None):
         empty = torch.empty(0, device=x.device)
-        y, save_mean, save_ivar, mask = ops._C.bn_fwd_train(
-            x, weight if weight is not None else empty,
-            bias if bias is not None else empty,
-            running_mean if running_mean is not None else empty,
-            running_var if running_var is not None else empty,
-            float(eps), float(momentum), bool(relu),
-            res if res is not None else empty)
+        if part is not None:
+            # stats partials already produced by the conv3x3_bn_fwd
+            # epilogue: ONE norm launch, no stats pass
+            y, save_mean, save_ivar, mask = ops._C.bn_fwd_train_part(
+                x, part, weight if weight is not None else empty,
+                bias if bias is not None else empty,
+                running_mean if running_mean is not None else empty,
+                running_var if running_var is not None else empty,
+                float(eps), float(momentum), bool(relu),
+                res if res is not None else empty)
+        else:
+            y, save_mean, save_ivar, mask = ops._C.bn_fwd_train(
+                x, weight if weight is not None else empty,
+                bias if bias is not None else empty,
+                running_mean if running_mean is not None else empty,
+                running_var if running_var is not None else empty,
+                float(eps), float(momentum), bool(relu),
+                res if res is not None else empty)
         ctx.relu = bool(relu)
         ctx.has_res = res is not None
         ctx.nhwc = _is_cl(x)
@@ -85,7 +96,7 @@ class _FusedBNFunction(torch.autograd.Function):
             weight if weight is not None else empty, ctx.relu, ctx.has_res)
         return (dx, dweight if weight is not None else None,
                 dbias if weight is not None else None, None, None, None,
-                None, None, dres if ctx.has_res else None)
+                None, None, dres if ctx.has_res else None, None)
 
 
 class FusedBatchNorm2d(nn.BatchNorm2d):
@@ -121,12 +132,15 @@ class FusedBatchNorm2d(nn.BatchNorm2d):
         else:
             xc = x.contiguous()
             rc = res.contiguous() if res is not None else None
+        part = getattr(x, '_ft_bn_part', None) if cl else None
+        if part is not None and not _nhwc_ok(self.num_features, x.dtype):
+            part = None
         return _FusedBNFunction.apply(
             xc, self.weight, self.bias,
             self.running_mean if self.track_running_stats else None,
             self.running_var if self.track_running_stats else None,
             momentum, self.eps,
-            self.fuse_relu or res is not None, rc)
+            self.fuse_relu or res is not None, rc, part)
 
     def _flush_nbt(self):
         if self._nbt_pending and self.num_batches_tracked is not None:

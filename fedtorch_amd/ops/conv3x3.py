@@ -28,6 +28,55 @@ _SHAPES = {(16, 32), (32, 16), (64, 8)}
 # profiles/r01_bench_notes.md "MFMA wrw" entry).  FEDTORCH_MFMA_WRW=1
 # re-enables for experimentation.
 _ENABLED = os.environ.get('FEDTORCH_MFMA_WRW', '0') == '1'
+# MFMA direct conv FORWARD with fused BN-stats epilogue (hip/convfwd.h):
+# 5.1-6.6 us/call vs MIOpen's solver stack AND the following BN's stats
+# pass dies (its partials come out of the conv epilogue).  Default ON.
+_FWD_ENABLED = os.environ.get('FEDTORCH_MFMA_FWD', '1') == '1'
+_EMPTY = {}
+
+
+def _empty(dev):
+    e = _EMPTY.get(dev)
+    if e is None:
+        e = _EMPTY[dev] = torch.empty(0, device=dev)
+    return e
+
+
+class _Conv3x3BNFn(torch.autograd.Function):
+    """y, part = conv3x3(x, w) with the BN-stats partials [grid, Co, 2]
+    produced by the conv epilogue (consumed by bn_fwd_train_part so the
+    following BN skips its stats pass).  Backward: dgrad/wrw via the same
+    paths as _Conv3x3Fn."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        grid = x.shape[0] * (x.shape[2] // 8) * \
+            (2 if weight.shape[0] == 64 else 1)
+        part = torch.empty(grid, weight.shape[0], 2, device=x.device)
+        e = _empty(x.device)
+        y = ops._C.conv3x3_bn_fwd(x, weight, part, e, e, e, False)
+        ctx.save_for_backward(x, weight)
+        ctx.mark_non_differentiable(part)
+        return y, part
+
+    @staticmethod
+    def backward(ctx, dy, _dpart):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=_CL)
+        if _ENABLED:
+            dx = None
+            if ctx.needs_input_grad[0]:
+                dx = torch.ops.aten.convolution_backward(
+                    dy, x, weight, None, [1, 1], [1, 1], [1, 1], False,
+                    [0, 0], 1, [True, False, False])[0]
+            dw = ops._C.conv3x3_wrw(dy, x)
+        else:
+            # ONE combined call (dgrad+wrw split into two was ~0.2 ms/step
+            # slower over the 19 body convs)
+            dx, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x, weight, None, [1, 1], [1, 1], [1, 1], False, [0, 0],
+                1, [bool(ctx.needs_input_grad[0]), True, False])
+        return dx, dw
 
 
 class _Conv3x3Fn(torch.autograd.Function):
@@ -56,6 +105,22 @@ class NhwcConv3x3(nn.Conv2d):
 
     def forward(self, x):
         w = self.weight
+        use_fwd = (_FWD_ENABLED and self.training and x.is_cuda
+                   and x.dim() == 4 and self.bias is None
+                   and self.stride == (1, 1) and self.padding == (1, 1)
+                   and x.dtype == torch.bfloat16
+                   and self.in_channels == self.out_channels
+                   and (self.out_channels, x.shape[3]) in _SHAPES
+                   and x.shape[2] % 8 == 0
+                   and x.is_contiguous(memory_format=_CL)
+                   and ops.hip_available() and not ops.FORCE_EAGER)
+        if use_fwd:
+            wb = w if w.dtype == torch.bfloat16 else \
+                w.bfloat16().contiguous(memory_format=_CL)
+            if wb.is_contiguous(memory_format=_CL):
+                y, part = _Conv3x3BNFn.apply(x, wb)
+                y._ft_bn_part = part  # consumed by FusedBatchNorm2d
+                return y
         use = (_ENABLED and torch.is_grad_enabled() and self.training
                and x.is_cuda
                and x.dim() == 4 and self.bias is None

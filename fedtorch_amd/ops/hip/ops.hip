@@ -1177,6 +1177,80 @@ static inline int bnh_ew_grid(long tasks) {
   return (int)(b < 1 ? 1 : b);
 }
 
+// collapse [G, C, 2] stats partials to [B2, C, 2]: bnh_norm_k's
+// per-block finalize re-reads ALL partial rows in EVERY normalize block
+// (B=1024 conv-epilogue rows measured 21 us/norm vs 7.9 at B~64).
+__global__ void __launch_bounds__(FT_BLOCK) part_reduce_k(
+    const float* __restrict__ part, long G, int cols /* C*2 */,
+    float* __restrict__ out, int B2) {
+  const long r0 = (long)blockIdx.x * G / B2;
+  const long r1 = (long)(blockIdx.x + 1) * G / B2;
+  for (int c = threadIdx.x; c < cols; c += FT_BLOCK) {
+    float s = 0.f;
+    for (long r = r0; r < r1; ++r) s += part[r * cols + c];
+    out[(long)blockIdx.x * cols + c] = s;
+  }
+}
+
+// bn_fwd_train with PRECOMPUTED stats partials (conv3x3_bn_fwd epilogue
+// rows, [B, C, 2] fp32): the bnh_stats pass is skipped entirely — one
+// bnh_norm_k launch finalizes the partials and normalizes.
+std::vector<torch::Tensor> bn_fwd_train_part(
+    torch::Tensor x, torch::Tensor part, torch::Tensor weight,
+    torch::Tensor bias, torch::Tensor running_mean, torch::Tensor running_var,
+    double eps, double momentum, bool relu, torch::Tensor res) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn_fwd_train_part: 4D GPU");
+  const bool nhwc =
+      x.is_contiguous(at::MemoryFormat::ChannelsLast) && !x.is_contiguous();
+  TORCH_CHECK(nhwc, "bn_fwd_train_part: channels_last only");
+  long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(part.is_cuda() && part.dim() == 3 && part.size(1) == C &&
+                  part.size(2) == 2 &&
+                  part.scalar_type() == torch::kFloat,
+              "bn_fwd_train_part: part must be [B, C, 2] fp32");
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto save_mean = torch::empty({C}, f32);
+  auto save_ivar = torch::empty({C}, f32);
+  auto y = torch::empty_like(x);
+  auto mask = torch::empty({0}, x.options().dtype(torch::kByte));
+  bool track = running_mean.numel() > 0;
+  if (part.size(0) > 64) {
+    const int B2 = 32;
+    auto small = torch::empty({B2, C, 2}, f32);
+    hipLaunchKernelGGL(part_reduce_k, dim3(B2), dim3(FT_BLOCK), 0, STREAM,
+                       part.data_ptr<float>(), part.size(0), (int)(C * 2),
+                       small.data_ptr<float>(), B2);
+    part = small;
+  }
+  const int B = (int)part.size(0);
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "bnh_fwd_part", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                 __hip_bfloat16, scalar_t>;
+    constexpr int VN = BnVec<T>::N;
+    TORCH_CHECK(bnh_ok(C, VN), "bn_fwd_train_part: unsupported C=", C);
+    const int lgc = bnh_lgc(C, VN);
+    const long NI = N * HW, tasks = NI << lgc;
+    hipLaunchKernelGGL((bnh_norm_k<T, typename BnVec<T>::V, VN>),
+                       dim3(bnh_ew_grid(tasks)), dim3(FT_BLOCK), 0, STREAM,
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       reinterpret_cast<T*>(y.data_ptr()),
+                       part.data_ptr<float>(), B,
+                       weight.numel() ? weight.data_ptr<float>() : nullptr,
+                       bias.numel() ? bias.data_ptr<float>() : nullptr,
+                       save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(),
+                       track ? running_mean.data_ptr<float>() : nullptr,
+                       track ? running_var.data_ptr<float>() : nullptr,
+                       res.numel()
+                           ? reinterpret_cast<const T*>(res.data_ptr())
+                           : nullptr,
+                       nullptr, NI, C, lgc, (float)eps, (float)momentum,
+                       relu ? 1 : 0);
+  });
+  return {y, save_mean, save_ivar, mask};
+}
+
 std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor bias,
                                         torch::Tensor running_mean,
@@ -1371,6 +1445,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train);
+  m.def("bn_fwd_train_part", &bn_fwd_train_part);
   m.def("bn_bwd", &bn_bwd);
   m.def("fused_sgd_step", &fused_sgd_step, "fused dual-mode SGD step");
   m.def("weighted_diff_restore", &weighted_diff_restore);

@@ -646,3 +646,66 @@ def test_fedadam_normalize_gpu_matches_eager():
         ops.FORCE_EAGER = False
     assert torch.allclose(v.cpu(), v_ref, atol=1e-5, rtol=1e-5)
     assert torch.allclose(g.cpu(), g_ref, atol=1e-5, rtol=1e-5)
+
+
+def test_conv3x3_bn_fused_fwd_matches_reference():
+    """conv3x3_bn_fwd + bn_fwd_train_part (stats from the conv epilogue)
+    must match the fp32 conv + BN reference on the same bf16 inputs."""
+    import torch.nn.functional as F
+    CL = torch.channels_last
+    torch.manual_seed(3)
+    for C, H in ((16, 32), (32, 16), (64, 8)):
+        N, W = 64, H
+        x = torch.randn(N, C, H, W, device='cuda').to(
+            memory_format=CL).bfloat16()
+        w = (torch.randn(C, C, 3, 3, device='cuda') / (3 * C) ** 0.5).to(
+            memory_format=CL).bfloat16()
+        gamma = torch.rand(C, device='cuda') + 0.5
+        beta = torch.randn(C, device='cuda') * 0.1
+        grid = N * (H // 8) * (2 if C == 64 else 1)
+        part = torch.empty(grid, C, 2, device='cuda')
+        e = torch.empty(0, device='cuda')
+        y = ops._C.conv3x3_bn_fwd(x, w, part, e, e, e, False)
+        z, sm, siv, _ = ops._C.bn_fwd_train_part(
+            y, part, gamma, beta, e, e, 1e-5, 0.1, True, e)
+        ref_y = F.conv2d(x.float(), w.float(), None, 1, 1)
+        ref = F.batch_norm(ref_y, None, None, gamma, beta, True, 0.1, 1e-5)
+        ref = torch.relu(ref)
+        err = (z.float() - ref).abs().max().item()
+        assert err < 0.12, 'C%d: fused conv+BN err %.4f' % (C, err)
+        m_ref = ref_y.mean(dim=(0, 2, 3))
+        assert torch.allclose(sm, m_ref, atol=2e-3), 'mean mismatch'
+
+
+def test_nhwc_conv_module_fused_path_autograd():
+    """NhwcConv3x3 -> BNReLU with the fused fwd + part handoff: full
+    fwd/bwd vs the eager fp32 module chain (same bf16 inputs)."""
+    import torch.nn.functional as F
+    from fedtorch_amd.ops.conv3x3 import NhwcConv3x3
+    from fedtorch_amd.ops.batchnorm import BNReLU, convert_to_fused_bn
+    CL = torch.channels_last
+    torch.manual_seed(4)
+    C, H = 16, 32
+    conv = NhwcConv3x3(C, C, kernel_size=3, stride=1, padding=1,
+                       bias=False).cuda().to(memory_format=CL)
+    bn = convert_to_fused_bn(torch.nn.Sequential(BNReLU(C))).cuda()
+    conv_w32 = conv.weight.detach().float().clone()
+    conv.weight.data = conv.weight.data.bfloat16()
+    conv.train(); bn.train()
+    x = torch.randn(64, C, H, H, device='cuda').to(
+        memory_format=CL).bfloat16().requires_grad_(True)
+    z = bn(conv(x))
+    loss = (z.float() ** 2).mean()
+    loss.backward()
+    # reference in fp32 on the same bf16 values
+    xr = x.detach().float().requires_grad_(True)
+    bnr = torch.nn.BatchNorm2d(C).cuda()
+    with torch.no_grad():
+        bnr.weight.copy_(bn[0].bn.weight)
+        bnr.bias.copy_(bn[0].bn.bias)
+    yr = F.conv2d(xr, conv_w32, None, 1, 1)
+    zr = torch.relu(bnr(yr))
+    lr_ = (zr ** 2).mean()
+    lr_.backward()
+    assert (z.float() - zr).abs().max().item() < 0.05
+    assert (x.grad.float() - xr.grad).abs().max().item() < 0.02
