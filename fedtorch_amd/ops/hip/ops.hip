@@ -1183,11 +1183,21 @@ static inline int bnh_ew_grid(long tasks) {
 __global__ void __launch_bounds__(FT_BLOCK) part_reduce_k(
     const float* __restrict__ part, long G, int cols /* C*2 */,
     float* __restrict__ out, int B2) {
+  // block b: rows [b*G/B2, (b+1)*G/B2), 2D thread split (row-parallel x
+  // col) + LDS tree — the first serial-per-column version measured
+  // 7.9 us/call (105 us/step over 19 layers)
+  __shared__ float lds[FT_BLOCK];
   const long r0 = (long)blockIdx.x * G / B2;
   const long r1 = (long)(blockIdx.x + 1) * G / B2;
-  for (int c = threadIdx.x; c < cols; c += FT_BLOCK) {
-    float s = 0.f;
-    for (long r = r0; r < r1; ++r) s += part[r * cols + c];
+  const int rp = FT_BLOCK / cols;            // row-parallel factor
+  const int c = threadIdx.x % cols, p = threadIdx.x / cols;
+  float s = 0.f;
+  if (p < rp)
+    for (long r = r0 + p; r < r1; r += rp) s += part[r * cols + c];
+  lds[threadIdx.x] = s;
+  __syncthreads();
+  if (p == 0) {
+    for (int i = 1; i < rp; ++i) s += lds[i * cols + c];
     out[(long)blockIdx.x * cols + c] = s;
   }
 }
