@@ -32,6 +32,9 @@ _ENABLED = os.environ.get('FEDTORCH_MFMA_WRW', '0') == '1'
 # 5.1-6.6 us/call vs MIOpen's solver stack AND the following BN's stats
 # pass dies (its partials come out of the conv epilogue).  Default ON.
 _FWD_ENABLED = os.environ.get('FEDTORCH_MFMA_FWD', '1') == '1'
+# MFMA wrw v2 (hip/convwrw2.h): transposed-LDS staging + alignbit tap
+# shifts; main kernel 12.8-15.5 us/call vs MIOpen's igemm+wrapper stack.
+_WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '1') == '1'
 _EMPTY = {}
 
 
@@ -63,13 +66,14 @@ class _Conv3x3BNFn(torch.autograd.Function):
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=_CL)
-        if _ENABLED:
+        if _WRW2_ENABLED or _ENABLED:
             dx = None
             if ctx.needs_input_grad[0]:
                 dx = torch.ops.aten.convolution_backward(
                     dy, x, weight, None, [1, 1], [1, 1], [1, 1], False,
                     [0, 0], 1, [True, False, False])[0]
-            dw = ops._C.conv3x3_wrw(dy, x)
+            dw = ops._C.conv3x3_wrw2(dy, x) if _WRW2_ENABLED \
+                else ops._C.conv3x3_wrw(dy, x)
         else:
             # ONE combined call (dgrad+wrw split into two was ~0.2 ms/step
             # slower over the 19 body convs)

@@ -15,6 +15,7 @@
 #include "batchnorm.h"
 #include "stemconv.h"
 #include "convwrw.h"
+#include "convwrw2.h"
 #include "convfwd.h"
 
 #define CHK(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on GPU")
@@ -863,6 +864,84 @@ void cast_to_half(torch::Tensor src, torch::Tensor dst) {
 }
 
 // ==========================================================================
+// MFMA wrw v2 (convwrw2.h): transposed LDS staging, alignbit tap shifts
+// ==========================================================================
+torch::Tensor conv3x3_wrw2(torch::Tensor dy, torch::Tensor x) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda() && dy.dim() == 4 && x.dim() == 4,
+              "conv3x3_wrw2: 4D GPU tensors");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast)
+                  && x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3_wrw2: channels_last only");
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16
+                  && x.scalar_type() == torch::kBFloat16,
+              "conv3x3_wrw2: bf16 only");
+  const int N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int Co = dy.size(1);
+  const bool ok = (Co == Ci) && ((Co == 16 && W == 32) ||
+                                 (Co == 32 && W == 16) ||
+                                 (Co == 64 && W == 8));
+  TORCH_CHECK(ok, "conv3x3_wrw2: unsupported (Co,Ci,W)=", Co, ",", Ci, ",",
+              W);
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto dw = torch::empty(
+      {Co, Ci, 3, 3},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const __hip_bfloat16* dyp =
+      reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr());
+  const __hip_bfloat16* xp =
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());
+  static const int cap = ft_env_int("FT_WRW2_NBLK", 1024);
+  int nblk, RQ, Q;
+  if (Co == 16) {
+    const long tiles = (long)N * (H / 4);        // SR=4
+    nblk = (int)(tiles < cap ? tiles : cap);
+    Q = 1;
+    RQ = nblk * 4;
+  } else if (Co == 32) {
+    const long tiles = (long)N * (H / 8);        // SR=8
+    nblk = (int)(tiles < cap ? tiles : cap);
+    Q = 4;
+    RQ = nblk;
+  } else {
+    const long tiles = (long)N;                  // SPOS = whole image
+    int streams = (int)(tiles < cap / 4 ? tiles : cap / 4);
+    nblk = streams * 4;
+    Q = 16;
+    RQ = streams;
+  }
+  auto part = torch::empty({(long)Q * RQ, 9L * 256}, f32);
+  auto acc = torch::zeros({(long)Q * 9 * 256}, f32);
+  if (Co == 16)
+    hipLaunchKernelGGL((conv3x3_wrw2_k<16, 16, 32>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H, RQ);
+  else if (Co == 32)
+    hipLaunchKernelGGL((conv3x3_wrw2_k<32, 32, 16>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H, RQ);
+  else
+    hipLaunchKernelGGL((conv3x3_wrw2_k<64, 64, 8>), dim3(nblk),
+                       dim3(FT_BLOCK), 0, STREAM, dyp, xp,
+                       part.data_ptr<float>(), N, H, RQ);
+  // stripes so the reduce grid covers the chip (~2048 blocks) without
+  // dropping below ~64 rows per block
+  int stripes = (int)(2048 / (Q * 36));
+  const int by_rows = RQ / 64;
+  if (stripes > by_rows) stripes = by_rows;
+  stripes = stripes < 1 ? 1 : (stripes > 16 ? 16 : stripes);
+  const int red_grid = Q * 36 * stripes;
+  hipLaunchKernelGGL(conv3x3_wrw2_reduce_k, dim3(red_grid), dim3(FT_BLOCK),
+                     0, STREAM, part.data_ptr<float>(), RQ, stripes,
+                     acc.data_ptr<float>());
+  hipLaunchKernelGGL(conv3x3_wrw2_cast_k,
+                     dim3((9 * Co * Ci + FT_BLOCK - 1) / FT_BLOCK),
+                     dim3(FT_BLOCK), 0, STREAM, acc.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dw.data_ptr()),
+                     Co, Ci);
+  return dw;
+}
+
+// ==========================================================================
 // MFMA direct NHWC 3x3/s1/p1 conv FORWARD with fused BN prologue/epilogue
 // (convfwd.h)
 // ==========================================================================
@@ -1479,5 +1558,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cast_to_half", &cast_to_half);
   m.def("mfma_probe", &mfma_probe);
   m.def("conv3x3_wrw", &conv3x3_wrw);
+  m.def("conv3x3_wrw2", &conv3x3_wrw2);
   m.def("conv3x3_bn_fwd", &conv3x3_bn_fwd);
 }

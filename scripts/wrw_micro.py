@@ -34,4 +34,20 @@ for C, W in [(16, 32), (32, 16), (64, 8)]:
         mi()
     e1.record(); torch.cuda.synchronize()
     miopen = e0.elapsed_time(e1) * 10
-    print('C=%-3d W=%-3d N=%d  mine %7.1f us   miopen %7.1f us' % (C, W, N, mine, miopen), flush=True)
+    # v2 correctness vs fp32 reference + timing
+    dw2 = ops._C.conv3x3_wrw2(dy, x)
+    ref = torch.ops.aten.convolution_backward(
+        dy.float(), x.float(), w.float(), None, [1, 1], [1, 1], [1, 1],
+        False, [0, 0], 1, [False, True, False])[1]
+    rel = (dw2.float() - ref).abs().max().item() / ref.abs().max().item()
+    for _ in range(10):
+        ops._C.conv3x3_wrw2(dy, x)
+    torch.cuda.synchronize()
+    e0.record()
+    for _ in range(100):
+        ops._C.conv3x3_wrw2(dy, x)
+    e1.record(); torch.cuda.synchronize()
+    v2 = e0.elapsed_time(e1) * 10
+    print('C=%-3d W=%-3d N=%d  v1 %7.1f us  v2 %7.1f us (rel err %.5f)  miopen %7.1f us'
+          % (C, W, N, mine, v2, rel, miopen), flush=True)
+    assert rel < 0.01, 'WRW2 numerics FAIL'
