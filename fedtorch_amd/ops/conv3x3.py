@@ -33,8 +33,12 @@ _ENABLED = os.environ.get('FEDTORCH_MFMA_WRW', '0') == '1'
 # pass dies (its partials come out of the conv epilogue).  Default ON.
 _FWD_ENABLED = os.environ.get('FEDTORCH_MFMA_FWD', '1') == '1'
 # MFMA wrw v2 (hip/convwrw2.h): transposed-LDS staging + alignbit tap
-# shifts; main kernel 12.8-15.5 us/call vs MIOpen's igemm+wrapper stack.
-_WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '1') == '1'
+# shifts.  Its MAIN kernel (12.8-15.5 us/call) beats MIOpen's igemm wrw,
+# but the partial-reduce/cast/zero auxiliary kernels per call still cost
+# more than they save in-bench (148.8k vs 163.5k samples/s flagship), so
+# default OFF until the partial pipeline is dieted.  FEDTORCH_WRW2=1 to
+# enable.
+_WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '0') == '1'
 _EMPTY = {}
 
 

@@ -709,3 +709,22 @@ def test_nhwc_conv_module_fused_path_autograd():
     lr_.backward()
     assert (z.float() - zr).abs().max().item() < 0.05
     assert (x.grad.float() - xr.grad).abs().max().item() < 0.02
+
+
+def test_conv3x3_wrw2_numerics():
+    """wrw v2 (transposed-LDS staging) vs the fp32 aten reference."""
+    CL = torch.channels_last
+    torch.manual_seed(9)
+    for C, W in ((16, 32), (32, 16), (64, 8)):
+        x = torch.randn(64, C, W, W, device='cuda').bfloat16().contiguous(
+            memory_format=CL)
+        dy = torch.randn(64, C, W, W, device='cuda').bfloat16().contiguous(
+            memory_format=CL)
+        w = torch.randn(C, C, 3, 3, device='cuda').bfloat16().contiguous(
+            memory_format=CL)
+        dw = ops._C.conv3x3_wrw2(dy, x)
+        ref = torch.ops.aten.convolution_backward(
+            dy.float(), x.float(), w.float(), None, [1, 1], [1, 1], [1, 1],
+            False, [0, 0], 1, [False, True, False])[1]
+        rel = (dw.float() - ref).abs().max().item() / ref.abs().max().item()
+        assert rel < 0.01, 'C%d rel %.5f' % (C, rel)
