@@ -91,11 +91,25 @@ def _make_loader(args, data, batch_size, shuffle, drop_last=False, tag=0):
     # client i exactly — the centered==distributed equivalence oracle
     #  depends on this, and it decouples loader order from
     # unrelated global-RNG consumption.
+    seed = int(args.manual_seed) * 1000003 + \
+        int(args.graph.rank) * 8191 + tag
+    if (getattr(args, 'device_data_cache', True) and args.graph.on_cuda
+            and torch.cuda.is_available()):
+        # MI355X-first: the whole partition lives in HBM and batches are
+        # tensor slices + batched device augmentation — the per-sample
+        # Python DataLoader path measured ~155 ms/step on the flagship
+        # parity loop vs 1.6 ms of GPU work (see datasets/device_cache.py)
+        from fedtorch_amd.components.datasets.device_cache import (
+            DeviceCachedLoader, NotCacheable)
+        try:
+            return DeviceCachedLoader(data, batch_size, seed=seed,
+                                      shuffle=shuffle, drop_last=drop_last)
+        except NotCacheable:
+            pass
     gen = None
     if shuffle:
         gen = torch.Generator()
-        gen.manual_seed(int(args.manual_seed) * 1000003 +
-                        int(args.graph.rank) * 8191 + tag)
+        gen.manual_seed(seed)
     return torch.utils.data.DataLoader(
         data, batch_size=batch_size, shuffle=shuffle, generator=gen,
         num_workers=args.num_workers, pin_memory=args.pin_memory,

@@ -199,6 +199,10 @@ def build_parser():
                         help='use the hand-written CDNA4 HIP kernel pack for '
                              'the arena hot paths (GPU only). When False, '
                              'fall back to eager torch ops.')
+    parser.add_argument('--device_data_cache', type=str2bool, default=True,
+                        help='keep each client partition resident in HBM '
+                             'and serve batches as tensor slices with '
+                             'batched on-GPU augmentation (GPU runs only)')
     parser.add_argument('--hip_graph', type=str2bool, default=False,
                         help='reserved. hipGraph capture of the local step '
                              'is implemented in the benchmark path '

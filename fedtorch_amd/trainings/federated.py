@@ -60,6 +60,10 @@ def run_local_steps(client, tracker, online=True, lambda_weight=None):
     lr = args.old_learning_rate
     if not online:
         return local_steps, lr
+    gs = getattr(client, 'graph_stepper', None)
+    if gs is None and getattr(args, 'hip_graph', False):
+        from fedtorch_amd.trainings.graphstep import GraphStepper
+        gs = client.graph_stepper = GraphStepper(client)
     is_sync = False
     while not is_sync:
         for _input, _target in client.train_loader:
@@ -77,6 +81,13 @@ def run_local_steps(client, tracker, online=True, lambda_weight=None):
                 # `main.py:104-106`)
                 is_sync = is_sync_fed(args)
                 break
+            if gs is not None and gs.maybe_step(_input, _target, lr):
+                # hipGraph replay did fwd/bwd/fused step; metrics land in
+                # the tracker at the sync flush
+                is_sync = is_sync_fed(args)
+                if is_sync:
+                    break
+                continue
             client.optimizer.zero_grad()
             with amp(args):
                 loss, performance = inference(
@@ -96,6 +107,8 @@ def run_local_steps(client, tracker, online=True, lambda_weight=None):
             is_sync = is_sync_fed(args)
             if is_sync:
                 break
+    if gs is not None:
+        gs.flush(tracker)
     return local_steps, lr
 
 
