@@ -61,6 +61,7 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_k(
   using C = Wrw2Cfg<CO, CI, W>;
   __shared__ __hip_bfloat16 s_dyT[CO * C::SD];
   __shared__ __hip_bfloat16 s_xT[CI * C::SX];
+  __shared__ float s_merge[C::Q == 1 ? 9 * 256 : 1];  // C16 wave merge
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE, lane = tid & (WAVE - 1);
@@ -162,12 +163,43 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_k(
   }
 
   // ---- partial row: quadrant-local [9][16][16] ---------------------------
-  // D of mfma(a=dy, b=x): col = ci-local = fm? NO: col = lane&15 maps the
-  // B operand's n index (ci), rows map A's m (co): row = kg*4 + reg.
-  const int rr = (C::Q == 1)
-                     ? (int)blockIdx.x * 4 + wave       // chunk partials
-                     : (C::Q == 4 ? q * RQ + (int)blockIdx.x
-                                  : q * RQ + (int)bstream);
+  // D of mfma(a=dy, b=x): col = lane&15 maps the B operand's n index
+  // (ci), rows map A's m (co): row = kg*4 + reg.
+  if (C::Q == 1) {
+    // the 4 waves hold CHUNK partials of the SAME quadrant: merge them
+    // through LDS so the block emits ONE row (4x less partial traffic —
+    // the reduce kernel was reading 37.7 MB/call for C16)
+    float* buf = s_merge;
+    for (int w = 1; w < 4; ++w) {
+      __syncthreads();
+      if (wave == w) {
+#pragma unroll
+        for (int t = 0; t < 9; ++t)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            buf[t * 256 + (kg * 4 + r) * 16 + fm] = acc[t][r];
+      }
+      __syncthreads();
+      if (wave == 0) {
+#pragma unroll
+        for (int t = 0; t < 9; ++t)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            acc[t][r] += buf[t * 256 + (kg * 4 + r) * 16 + fm];
+      }
+    }
+    if (wave == 0) {
+      float* pr = part + (long)blockIdx.x * (9 * 16 * 16);
+#pragma unroll
+      for (int t = 0; t < 9; ++t)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          pr[t * 256 + (kg * 4 + r) * 16 + fm] = acc[t][r];
+    }
+    return;
+  }
+  const int rr = (C::Q == 4) ? q * RQ + (int)blockIdx.x
+                             : q * RQ + (int)bstream;
   float* pr = part + (long)rr * (9 * 16 * 16);
 #pragma unroll
   for (int t = 0; t < 9; ++t) {
@@ -178,13 +210,12 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_k(
   }
 }
 
-// part[q*RQ + r][9*16*16] -> fp32 accumulator (stripe-parallel: the one-
-// kernel 36-block version was latency-bound — 36 blocks x up to 2048
-// serial 256 B row visits on an otherwise idle chip).  Each block reduces
-// <= 256 rows of 64 quadrant-local columns and atomicAdds once per column.
+// part[q*RQ + r][9*16*16] -> [stripe][Q*9*256] (stripe-parallel, no
+// atomics — the atomic+acc-zero version cost a fill kernel + atomic
+// round per call); conv3x3_wrw2_cast_k folds the stripes and casts.
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_reduce_k(
     const float* __restrict__ part, int RQ, int stripes,
-    float* __restrict__ acc /* [Q*9*256] quadrant-local */) {
+    float* __restrict__ out /* [stripes][Q*9*256] */, int Q) {
   __shared__ float lds[4][64];
   const int c = threadIdx.x & 63, rs = threadIdx.x >> 6;
   const int blocks_per_q = (9 * 256) / 64 * stripes;
@@ -201,25 +232,25 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_reduce_k(
   __syncthreads();
   if (rs == 0) {
     s = lds[0][c] + lds[1][c] + lds[2][c] + lds[3][c];
-    if (stripes == 1)
-      acc[(long)q * (9 * 256) + lc] = s;
-    else
-      atomicAdd(&acc[(long)q * (9 * 256) + lc], s);
+    out[(long)stripe * (Q * 9 * 256) + (long)q * (9 * 256) + lc] = s;
   }
 }
 
-// quadrant-local fp32 acc -> dw bf16 [co][kh][kw][ci] (channels_last)
+// fold stripes + cast: [stripes][Q*9*256] fp32 -> dw bf16 [co][kh][kw][ci]
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_wrw2_cast_k(
-    const float* __restrict__ acc, __hip_bfloat16* __restrict__ dw,
-    int CO, int CI) {
+    const float* __restrict__ out, int stripes,
+    __hip_bfloat16* __restrict__ dw, int CO, int CI) {
   const int total = 9 * CO * CI;
+  const int QC = ((CO / 16) * (CI / 16)) * 9 * 256;
   for (int e = blockIdx.x * FT_BLOCK + threadIdx.x; e < total;
        e += gridDim.x * FT_BLOCK) {
     const int q = e / (9 * 256), lc = e % (9 * 256);
+    float s = 0.f;
+    for (int st = 0; st < stripes; ++st)
+      s += out[(long)st * QC + (long)q * (9 * 256) + lc];
     const int tap = lc / 256, mrow = (lc % 256) / 16, ncol = lc % 16;
     const int co = (q / (CI / 16)) * 16 + mrow;
     const int ci = (q % (CI / 16)) * 16 + ncol;
-    dw[((long)co * 9 + tap) * CI + ci] =
-        __float2bfloat16(acc[(long)q * (9 * 256) + lc]);
+    dw[((long)co * 9 + tap) * CI + ci] = __float2bfloat16(s);
   }
 }

@@ -896,7 +896,7 @@ torch::Tensor conv3x3_wrw2(torch::Tensor dy, torch::Tensor x) {
     const long tiles = (long)N * (H / 4);        // SR=4
     nblk = (int)(tiles < cap ? tiles : cap);
     Q = 1;
-    RQ = nblk * 4;
+    RQ = nblk;                                   // waves merged in-block
   } else if (Co == 32) {
     const long tiles = (long)N * (H / 8);        // SR=8
     nblk = (int)(tiles < cap ? tiles : cap);
@@ -904,13 +904,12 @@ torch::Tensor conv3x3_wrw2(torch::Tensor dy, torch::Tensor x) {
     RQ = nblk;
   } else {
     const long tiles = (long)N;                  // SPOS = whole image
-    int streams = (int)(tiles < cap / 4 ? tiles : cap / 4);
+    int streams = (int)(tiles < cap / 8 ? tiles : cap / 8);
     nblk = streams * 4;
     Q = 16;
     RQ = streams;
   }
   auto part = torch::empty({(long)Q * RQ, 9L * 256}, f32);
-  auto acc = torch::zeros({(long)Q * 9 * 256}, f32);
   if (Co == 16)
     hipLaunchKernelGGL((conv3x3_wrw2_k<16, 16, 32>), dim3(nblk),
                        dim3(FT_BLOCK), 0, STREAM, dyp, xp,
@@ -929,13 +928,15 @@ torch::Tensor conv3x3_wrw2(torch::Tensor dy, torch::Tensor x) {
   const int by_rows = RQ / 64;
   if (stripes > by_rows) stripes = by_rows;
   stripes = stripes < 1 ? 1 : (stripes > 16 ? 16 : stripes);
+  auto red = torch::empty({(long)stripes * Q * 9 * 256}, f32);
   const int red_grid = Q * 36 * stripes;
   hipLaunchKernelGGL(conv3x3_wrw2_reduce_k, dim3(red_grid), dim3(FT_BLOCK),
                      0, STREAM, part.data_ptr<float>(), RQ, stripes,
-                     acc.data_ptr<float>());
+                     red.data_ptr<float>(), Q);
   hipLaunchKernelGGL(conv3x3_wrw2_cast_k,
                      dim3((9 * Co * Ci + FT_BLOCK - 1) / FT_BLOCK),
-                     dim3(FT_BLOCK), 0, STREAM, acc.data_ptr<float>(),
+                     dim3(FT_BLOCK), 0, STREAM, red.data_ptr<float>(),
+                     stripes,
                      reinterpret_cast<__hip_bfloat16*>(dw.data_ptr()),
                      Co, Ci);
   return dw;
