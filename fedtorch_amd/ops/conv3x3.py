@@ -39,6 +39,9 @@ _FWD_ENABLED = os.environ.get('FEDTORCH_MFMA_FWD', '1') == '1'
 # default OFF until the partial pipeline is dieted.  FEDTORCH_WRW2=1 to
 # enable.
 _WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '0') == '1'
+# MFMA direct conv backward-data (hip/convfwd.h conv3x3_dgrad_k): the fwd
+# kernel's mirror, 5.0-6.7 us/call vs MIOpen's 23-24.  Default ON.
+_DGRAD_ENABLED = os.environ.get('FEDTORCH_MFMA_DGRAD', '1') == '1'
 _EMPTY = {}
 
 
@@ -70,7 +73,20 @@ class _Conv3x3BNFn(torch.autograd.Function):
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=_CL)
-        if _WRW2_ENABLED or _ENABLED:
+        custom_dx = _DGRAD_ENABLED
+        custom_dw = _WRW2_ENABLED or _ENABLED
+        if custom_dx and custom_dw:
+            dx = ops._C.conv3x3_dgrad(dy, weight) \
+                if ctx.needs_input_grad[0] else None
+            dw = ops._C.conv3x3_wrw2(dy, x) if _WRW2_ENABLED \
+                else ops._C.conv3x3_wrw(dy, x)
+        elif custom_dx:
+            dx = ops._C.conv3x3_dgrad(dy, weight) \
+                if ctx.needs_input_grad[0] else None
+            dw = torch.ops.aten.convolution_backward(
+                dy, x, weight, None, [1, 1], [1, 1], [1, 1], False, [0, 0],
+                1, [False, True, False])[1]
+        elif custom_dw:
             dx = None
             if ctx.needs_input_grad[0]:
                 dx = torch.ops.aten.convolution_backward(

@@ -240,3 +240,142 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
     }
   }
 }
+
+
+// ==========================================================================
+// MFMA direct NHWC 3x3/s1/p1 conv BACKWARD-DATA (gfx950) — the fwd
+// kernel's mirror: dx[p, ci] = sum_{tap, co} dy[p - shift(tap)][co] *
+// w[co][tap][ci].  GEMM: A[m=pixel][k=(tap, co)] = dy slab reads
+// (channels contiguous, same staging as fwd's x), B[k][n=ci] = weights
+// gathered as [k/8][CI_TILE][8] with the tap REVERSED (full correlation
+// <-> convolution flip), K = 9*CO resident in LDS.
+// ==========================================================================
+template <int CI, int CO, int CI_TILE, int W>
+__global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
+    const __hip_bfloat16* __restrict__ dy,  // [N, H, W, CO]
+    const __hip_bfloat16* __restrict__ w,   // [CO, 3, 3, CI]
+    __hip_bfloat16* __restrict__ dx,        // [N, H, W, CI]
+    int N, int H) {
+  constexpr int R = 8;
+  constexpr int KTOT = 9 * CO;
+  constexpr int KPAD = (KTOT + 31) & ~31;
+  constexpr int K8 = KPAD / 8;
+  constexpr int COP = CO + 8;
+  constexpr int XR = R + 2, XC = W + 2;
+  constexpr int NSLOT = XR * XC + 1;
+  constexpr int P = R * W;
+  constexpr int MT = P / 16;
+  constexpr int MTW = MT / 4;
+  constexpr int NT = CI_TILE / 16;
+
+  __shared__ __hip_bfloat16 ys[NSLOT * COP];     // dy slab (+halo, pad)
+  __shared__ __hip_bfloat16 bw[K8 * CI_TILE * 8];
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE, lane = tid & (WAVE - 1);
+  const int fm = lane & 15, kg = lane >> 4;
+
+  const int cib = blockIdx.x % (CI / CI_TILE);
+  const int rb = (blockIdx.x / (CI / CI_TILE)) % (H / R);
+  const int n = blockIdx.x / ((CI / CI_TILE) * (H / R));
+  const int ci0 = cib * CI_TILE;
+  const int h0 = rb * R;
+
+  // ---- stage weights: bw[k/8][ci][j] = w[co][2-dh][2-dw][ci0+ci],
+  //      k = tap*CO + co (co runs of 8 within one tap: CO % 8 == 0) ----
+  for (int e = tid; e < K8 * CI_TILE; e += FT_BLOCK) {
+    const int k8 = e / CI_TILE, ci = e % CI_TILE;
+    const int k = k8 * 8;
+    __hip_bfloat16 vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_bfloat16 v; *reinterpret_cast<unsigned short*>(&v) = 0;
+      if (k + j < KTOT) {
+        const int tap = (k + j) / CO, co = (k + j) % CO;
+        const int rev = 8 - tap;                   // (2-dh)*3 + (2-dw)
+        v = w[((long)co * 9 + rev) * CI + ci0 + ci];
+      }
+      vals[j] = v;
+    }
+    *reinterpret_cast<uint4*>(&bw[((long)k8 * CI_TILE + ci) * 8]) =
+        *reinterpret_cast<const uint4*>(vals);
+  }
+
+  // ---- stage dy slab (identical structure to the fwd x slab) ----------
+  for (int e = tid; e < NSLOT; e += FT_BLOCK)
+    *reinterpret_cast<uint4*>(&ys[(long)e * COP + CO]) = uint4{0, 0, 0, 0};
+  for (int e = tid; e < XR * 2 + (COP / 8); e += FT_BLOCK) {
+    if (e < XR * 2) {
+      const int row = e >> 1, col = (e & 1) ? (XC - 1) : 0;
+      __hip_bfloat16* p = &ys[(long)(row * XC + col) * COP];
+      for (int j = 0; j < CO; j += 8)
+        *reinterpret_cast<uint4*>(p + j) = uint4{0, 0, 0, 0};
+    } else {
+      *reinterpret_cast<uint4*>(
+          &ys[(long)(NSLOT - 1) * COP + (e - XR * 2) * 8]) =
+          uint4{0, 0, 0, 0};
+    }
+  }
+  __syncthreads();
+  constexpr int BCH = W * CO / 8;
+  for (int e = tid; e < XR * BCH; e += FT_BLOCK) {
+    const int row = e / BCH, c = e % BCH;
+    const int hh = h0 - 1 + row;
+    uint4 v = {0, 0, 0, 0};
+    if (hh >= 0 && hh < H)
+      v = *reinterpret_cast<const uint4*>(
+          dy + (((long)n * H + hh) * W) * CO + c * 8);
+    const int pix = c * 8 / CO, co8 = c * 8 % CO;
+    *reinterpret_cast<uint4*>(
+        &ys[(long)(row * XC + 1 + pix) * COP + co8]) = v;
+  }
+  __syncthreads();
+
+  // ---- MFMAs ----------------------------------------------------------
+  f32x4 acc[MTW][NT];
+#pragma unroll
+  for (int mt = 0; mt < MTW; ++mt)
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+  for (int ks = 0; ks < KPAD / 32; ++ks) {
+    const int kk = ks * 32 + kg * 8;
+    const int tap = kk / CO, co8 = kk % CO;
+    // dx[p] needs dy[p + (dh-1, dw-1)] for REVERSED w tap; with tap
+    // enumerating the REVERSED kernel, the dy offset is (dh, dw) of the
+    // forward halo walk: same slot arithmetic as the fwd kernel.
+    const int dh = tap / 3, dw = tap % 3;
+#pragma unroll
+    for (int mt = 0; mt < MTW; ++mt) {
+      const int p = (wave * MTW + mt) * 16 + fm;
+      const int r = p / W, c = p % W;
+      const int slot = (tap < 9) ? ((r + dh) * XC + (c + dw)) : (NSLOT - 1);
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &ys[(long)slot * COP + co8]);
+#pragma unroll
+      for (int nt = 0; nt < NT; ++nt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &bw[((long)(ks * 4 + kg) * CI_TILE + nt * 16 + fm) * 8]);
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, acc[mt][nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- store dx --------------------------------------------------------
+#pragma unroll
+  for (int mt = 0; mt < MTW; ++mt) {
+    const int ptile = (wave * MTW + mt) * 16;
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+#pragma unroll
+      for (int r4 = 0; r4 < 4; ++r4) {
+        const int p = ptile + kg * 4 + r4;
+        const int hh = h0 + p / W, cc = p % W;
+        dx[(((long)n * H + hh) * W + cc) * CI + ci0 + nt * 16 + fm] =
+            (__hip_bfloat16)acc[mt][nt][r4];
+      }
+    }
+  }
+}

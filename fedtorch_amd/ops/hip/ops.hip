@@ -1026,6 +1026,49 @@ torch::Tensor conv3x3_bn_fwd(torch::Tensor x, torch::Tensor w,
 }
 
 // ==========================================================================
+// MFMA direct conv backward-data (convfwd.h conv3x3_dgrad_k)
+// ==========================================================================
+torch::Tensor conv3x3_dgrad(torch::Tensor dy, torch::Tensor w) {
+  TORCH_CHECK(dy.is_cuda() && w.is_cuda() && dy.dim() == 4,
+              "conv3x3_dgrad: 4D GPU tensors");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3_dgrad: channels_last only");
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 &&
+                  w.scalar_type() == torch::kBFloat16,
+              "conv3x3_dgrad: bf16 only");
+  const int N = dy.size(0), Co = dy.size(1), H = dy.size(2),
+            W = dy.size(3);
+  const int Ci = w.size(1);
+  const bool ok = (Co == Ci) && ((Co == 16 && W == 32) ||
+                                 (Co == 32 && W == 16) ||
+                                 (Co == 64 && W == 8));
+  TORCH_CHECK(ok, "conv3x3_dgrad: unsupported (Co,Ci,W)=", Co, ",", Ci,
+              ",", W);
+  TORCH_CHECK(H % 8 == 0, "conv3x3_dgrad: H % 8 != 0");
+  auto dx = torch::empty(
+      {N, Ci, H, W},
+      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const __hip_bfloat16* dyp =
+      reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr());
+  const __hip_bfloat16* wp =
+      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
+  __hip_bfloat16* dxp = reinterpret_cast<__hip_bfloat16*>(dx.data_ptr());
+  if (Co == 16)
+    hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32>),
+                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
+                       wp, dxp, N, H);
+  else if (Co == 32)
+    hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16>),
+                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
+                       wp, dxp, N, H);
+  else
+    hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8>),
+                       dim3(N * (H / 8) * 2), dim3(FT_BLOCK), 0, STREAM,
+                       dyp, wp, dxp, N, H);
+  return dx;
+}
+
+// ==========================================================================
 // MFMA 3x3/s1/p1 NHWC bf16 conv weight gradient (convwrw.h)
 // ==========================================================================
 torch::Tensor conv3x3_wrw(torch::Tensor dy, torch::Tensor x) {
@@ -1560,5 +1603,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("conv3x3_wrw", &conv3x3_wrw);
   m.def("conv3x3_wrw2", &conv3x3_wrw2);
+  m.def("conv3x3_dgrad", &conv3x3_dgrad);
   m.def("conv3x3_bn_fwd", &conv3x3_bn_fwd);
 }

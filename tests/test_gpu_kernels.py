@@ -728,3 +728,22 @@ def test_conv3x3_wrw2_numerics():
             False, [0, 0], 1, [False, True, False])[1]
         rel = (dw.float() - ref).abs().max().item() / ref.abs().max().item()
         assert rel < 0.01, 'C%d rel %.5f' % (C, rel)
+
+
+def test_conv3x3_dgrad_numerics():
+    """custom backward-data (fwd kernel's mirror) vs fp32 aten."""
+    CL = torch.channels_last
+    torch.manual_seed(13)
+    for C, W in ((16, 32), (32, 16), (64, 8)):
+        dy = torch.randn(64, C, W, W, device='cuda').bfloat16().contiguous(
+            memory_format=CL)
+        x = torch.randn(64, C, W, W, device='cuda').bfloat16().contiguous(
+            memory_format=CL)
+        w = (torch.randn(C, C, 3, 3, device='cuda') / C).bfloat16()
+        w = w.contiguous(memory_format=CL)
+        dx = ops._C.conv3x3_dgrad(dy, w)
+        ref = torch.ops.aten.convolution_backward(
+            dy.float(), x.float(), w.float(), None, [1, 1], [1, 1], [1, 1],
+            False, [0, 0], 1, [True, False, False])[0]
+        rel = (dx.float() - ref).abs().max().item() / ref.abs().max().item()
+        assert rel < 0.01, 'C%d rel %.5f' % (C, rel)
