@@ -64,6 +64,7 @@ class _FusedBNFunction(torch.autograd.Function):
         ctx.has_res = res is not None
         ctx.nhwc = _is_cl(x)
         ctx.has_mask = mask.numel() > 0
+        ctx.from_fused_conv = part is not None
         if ctx.has_mask:
             # ReLU bitmask replaces the saved output (bwd reads 1 bit/elem)
             ctx.save_for_backward(x, weight, save_mean, save_ivar, mask)
@@ -89,6 +90,24 @@ class _FusedBNFunction(torch.autograd.Function):
         else:
             dy = dy.contiguous()
         empty = torch.empty(0, device=x.device)
+        from fedtorch_amd.ops import conv3x3 as _c3
+        if (ctx.from_fused_conv and ctx.nhwc and not ctx.has_mask
+                and _c3.bn_defer_active()):
+            # deferred dx: the producing conv's backward applies the
+            # per-channel affine transform while staging (and emits the
+            # transformed dy for its wrw) — return dz tagged via the
+            # side table instead of running bnh_bwd_dx here
+            z = y if ctx.relu else None
+            coefs, dweight, dbias, dres = ops._C.bn_bwd_defer(
+                dy, x, z if z is not None else empty,
+                save_mean, save_ivar,
+                weight if weight is not None else empty,
+                ctx.relu, ctx.has_res)
+            _c3.register_bn_defer(dy, x, z, coefs, ctx.relu)
+            return (dy, dweight if weight is not None else None,
+                    dbias if weight is not None else None, None, None,
+                    None, None, None,
+                    dres if ctx.has_res else None, None)
         dx, dweight, dbias, dres = ops._C.bn_bwd(
             dy, x, y, mask if mask is not None else
             torch.empty(0, device=x.device, dtype=torch.uint8),

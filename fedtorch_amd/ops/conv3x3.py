@@ -42,6 +42,27 @@ _WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '0') == '1'
 # MFMA direct conv backward-data (hip/convfwd.h conv3x3_dgrad_k): the fwd
 # kernel's mirror, 5.0-6.7 us/call vs MIOpen's 23-24.  Default ON.
 _DGRAD_ENABLED = os.environ.get('FEDTORCH_MFMA_DGRAD', '1') == '1'
+# Deferred BN backward (hip/convfwd.h TRF path): the BN's elementwise dx
+# pass runs inside the conv dgrad staging (which also writes the
+# transformed dy for the wrw consumer) — the bnh_bwd_dx kernel leaves the
+# hot path.  Handshake: _FusedBNFunction.backward registers
+# (xbn, z, coefs, relu) under its outgoing grad tensor's data_ptr; the
+# producing conv's backward pops it (python tensor attrs do not survive
+# the autograd engine's rewrapping, a data_ptr-keyed side table does).
+# MEASURED A WASH at the flagship config (174.0k vs 174.7k samples/s):
+# the killed bnh_bwd_dx pass (~145 us/step) is consumed by the coef/mask
+# side kernels + the 2 extra tensors the transform stages; default OFF,
+# capability kept (tests pin deferred == eager gradients).
+_BNDEFER_ENABLED = os.environ.get('FEDTORCH_BN_DEFER', '0') == '1'
+_BNBWD_TAGS = {}
+
+
+def bn_defer_active():
+    return _DGRAD_ENABLED and _BNDEFER_ENABLED and not ops.FORCE_EAGER
+
+
+def register_bn_defer(grad, xbn, z, coefs, relu):
+    _BNBWD_TAGS[grad.data_ptr()] = (xbn, z, coefs, relu)
 _EMPTY = {}
 
 
@@ -73,6 +94,20 @@ class _Conv3x3BNFn(torch.autograd.Function):
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=_CL)
+        tag = _BNBWD_TAGS.pop(dy.data_ptr(), None) if _BNBWD_TAGS else None
+        if tag is not None and _DGRAD_ENABLED:
+            xbn, z, coefs, relu = tag
+            dx, dyc = ops._C.conv3x3_dgrad_bn(
+                dy, weight, xbn, z if z is not None else xbn, coefs, relu)
+            if not ctx.needs_input_grad[0]:
+                dx = None
+            if _WRW2_ENABLED:
+                dw = ops._C.conv3x3_wrw2(dyc, x)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    dyc, x, weight, None, [1, 1], [1, 1], [1, 1], False,
+                    [0, 0], 1, [False, True, False])[1]
+            return dx, dw
         custom_dx = _DGRAD_ENABLED
         custom_dw = _WRW2_ENABLED or _ENABLED
         if custom_dx and custom_dw:

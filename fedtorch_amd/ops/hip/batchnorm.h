@@ -601,3 +601,69 @@ __global__ void bn_bwd_dx_k(const T* __restrict__ dy, const T* __restrict__ x,
     if (dres) *reinterpret_cast<VT*>(dres + base) = gr;
   }
 }
+
+
+// finalize bnh_bwd_stats partials into the per-channel AFFINE transform
+// of the deferred BN backward (consumed by conv3x3_dgrad_k<TRF=true>):
+//   dx_bn = A[c]*g_masked + B[c] + D[c]*x
+// with A = gamma*ivar, B = A*(-mdy + mean*ivar*mdyxh), D = -A*ivar*mdyxh
+// (algebraic rearrangement of bnh_bwd_dx_k's formula).  Also emits
+// dgamma = sum(dy*xhat), dbeta = sum(dy).
+__global__ void __launch_bounds__(FT_BLOCK) bnh_bwd_coef_k(
+    const float* __restrict__ part, int B, long NI,
+    const float* __restrict__ save_mean,
+    const float* __restrict__ save_ivar,
+    const float* __restrict__ weight, long C,
+    float* __restrict__ coefs /* [3, C] */, float* __restrict__ dweight,
+    float* __restrict__ dbias) {
+  __shared__ float fa[FT_BLOCK], fb[FT_BLOCK];
+  const float inv_n = 1.0f / (float)NI;
+  const int P = blockDim.x / (int)C;
+  const int c0 = threadIdx.x % (int)C, p0 = threadIdx.x / (int)C;
+  float a = 0.f, bb = 0.f;
+  for (int i = p0; i < B; i += P) {
+    const float* p = part + ((long)i * C + c0) * 2;
+    a += p[0];
+    bb += p[1];
+  }
+  fa[threadIdx.x] = a;
+  fb[threadIdx.x] = bb;
+  __syncthreads();
+  const int c = threadIdx.x;
+  if (c < C) {
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = 0; i < P; ++i) {
+      s1 += fa[i * C + c];
+      s2 += fb[i * C + c];
+    }
+    const float w = weight ? weight[c] : 1.f;
+    const float iv = save_ivar[c];
+    const float A = w * iv;
+    const float mdy = s1 * inv_n, mdyxh = s2 * inv_n;
+    coefs[c] = A;
+    coefs[C + c] = A * (fmaf(save_mean[c] * iv, mdyxh, -mdy));
+    coefs[2 * C + c] = -A * iv * mdyxh;
+    dweight[c] = s2;
+    dbias[c] = s1;
+  }
+}
+
+// dres for the deferred BN backward: the ReLU-masked upstream gradient
+// (the fused residual branch needs it immediately; the conv transform
+// recomputes the same mask internally for its own staging)
+template <typename T, typename VT, int VN>
+__global__ void __launch_bounds__(FT_BLOCK) bnh_mask_dres_k(
+    const T* __restrict__ dy, const T* __restrict__ yv,
+    T* __restrict__ dres, long total_v) {
+  const unsigned stride = gridDim.x * blockDim.x;
+  for (unsigned t = blockIdx.x * blockDim.x + threadIdx.x; t < total_v;
+       t += stride) {
+    const long base = (long)t * VN;
+    VT g = *reinterpret_cast<const VT*>(dy + base);
+    VT yy = *reinterpret_cast<const VT*>(yv + base);
+#pragma unroll
+    for (int j = 0; j < VN; ++j)
+      if ((float)yy.d[j] <= 0.f) g.d[j] = (T)0.f;
+    *reinterpret_cast<VT*>(dres + base) = g;
+  }
+}

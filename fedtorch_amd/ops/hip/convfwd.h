@@ -250,12 +250,16 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
 // gathered as [k/8][CI_TILE][8] with the tap REVERSED (full correlation
 // <-> convolution flip), K = 9*CO resident in LDS.
 // ==========================================================================
-template <int CI, int CO, int CI_TILE, int W>
+template <int CI, int CO, int CI_TILE, int W, bool TRF>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
-    const __hip_bfloat16* __restrict__ dy,  // [N, H, W, CO]
+    const __hip_bfloat16* __restrict__ dy,  // [N, H, W, CO] (dz when TRF)
     const __hip_bfloat16* __restrict__ w,   // [CO, 3, 3, CI]
     __hip_bfloat16* __restrict__ dx,        // [N, H, W, CI]
-    int N, int H) {
+    const __hip_bfloat16* __restrict__ xbn, // conv output (BN input)
+    const __hip_bfloat16* __restrict__ zv,  // BN output (ReLU mask src)
+    const float* __restrict__ coefs,        // [3, CO] A, B, D
+    __hip_bfloat16* __restrict__ dyc,       // transformed dy out (for wrw)
+    int N, int H, int relu) {
   constexpr int R = 8;
   constexpr int KTOT = 9 * CO;
   constexpr int KPAD = (KTOT + 31) & ~31;
@@ -270,6 +274,7 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
 
   __shared__ __hip_bfloat16 ys[NSLOT * COP];     // dy slab (+halo, pad)
   __shared__ __hip_bfloat16 bw[K8 * CI_TILE * 8];
+  __shared__ float sA[TRF ? CO : 1], sB[TRF ? CO : 1], sD[TRF ? CO : 1];
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE, lane = tid & (WAVE - 1);
@@ -280,6 +285,12 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
   const int n = blockIdx.x / ((CI / CI_TILE) * (H / R));
   const int ci0 = cib * CI_TILE;
   const int h0 = rb * R;
+
+  if (TRF && tid < CO) {
+    sA[tid] = coefs[tid];
+    sB[tid] = coefs[CO + tid];
+    sD[tid] = coefs[2 * CO + tid];
+  }
 
   // ---- stage weights: bw[k/8][ci][j] = w[co][2-dh][2-dw][ci0+ci],
   //      k = tap*CO + co (co runs of 8 within one tap: CO % 8 == 0) ----
@@ -322,10 +333,33 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
     const int row = e / BCH, c = e % BCH;
     const int hh = h0 - 1 + row;
     uint4 v = {0, 0, 0, 0};
+    const long gbase = (((long)n * H + hh) * W) * CO + c * 8;
     if (hh >= 0 && hh < H)
-      v = *reinterpret_cast<const uint4*>(
-          dy + (((long)n * H + hh) * W) * CO + c * 8);
+      v = *reinterpret_cast<const uint4*>(dy + gbase);
     const int pix = c * 8 / CO, co8 = c * 8 % CO;
+    if (TRF && hh >= 0 && hh < H) {
+      // deferred BN backward: dy_conv = A*relu_mask(dz) + B + D*x_bn,
+      // staged AND written back out for the weight-gradient consumer
+      const uint4 xv4 = *reinterpret_cast<const uint4*>(xbn + gbase);
+      const uint4 zv4 = *reinterpret_cast<const uint4*>(zv + gbase);
+      const __hip_bfloat16* gg =
+          reinterpret_cast<const __hip_bfloat16*>(&v);
+      const __hip_bfloat16* xx =
+          reinterpret_cast<const __hip_bfloat16*>(&xv4);
+      const __hip_bfloat16* zz =
+          reinterpret_cast<const __hip_bfloat16*>(&zv4);
+      __hip_bfloat16 out[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = (float)gg[j];
+        if (relu && (float)zz[j] <= 0.f) g = 0.f;
+        out[j] = (__hip_bfloat16)(
+            fmaf(sA[co8 + j], g,
+                 fmaf(sD[co8 + j], (float)xx[j], sB[co8 + j])));
+      }
+      v = *reinterpret_cast<const uint4*>(out);
+      *reinterpret_cast<uint4*>(dyc + gbase) = v;
+    }
     *reinterpret_cast<uint4*>(
         &ys[(long)(row * XC + 1 + pix) * COP + co8]) = v;
   }

@@ -1054,19 +1054,78 @@ torch::Tensor conv3x3_dgrad(torch::Tensor dy, torch::Tensor w) {
       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
   __hip_bfloat16* dxp = reinterpret_cast<__hip_bfloat16*>(dx.data_ptr());
   if (Co == 16)
-    hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32>),
+    hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32, false>),
                        dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
-                       wp, dxp, N, H);
+                       wp, dxp, nullptr, nullptr, nullptr, nullptr, N, H,
+                       0);
   else if (Co == 32)
-    hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16>),
+    hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16, false>),
                        dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
-                       wp, dxp, N, H);
+                       wp, dxp, nullptr, nullptr, nullptr, nullptr, N, H,
+                       0);
   else
-    hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8>),
+    hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8, false>),
                        dim3(N * (H / 8) * 2), dim3(FT_BLOCK), 0, STREAM,
-                       dyp, wp, dxp, N, H);
+                       dyp, wp, dxp, nullptr, nullptr, nullptr, nullptr,
+                       N, H, 0);
   return dx;
 }
+
+// deferred-BN-backward dgrad: dy here is dz (the BN output gradient);
+// the kernel applies dy_conv = A*mask(dz) + B + D*x_bn while staging and
+// ALSO writes dy_conv out (the wrw consumer reads it) — the separate
+// bnh_bwd_dx pass disappears.
+std::vector<torch::Tensor> conv3x3_dgrad_bn(
+    torch::Tensor dz, torch::Tensor w, torch::Tensor xbn, torch::Tensor z,
+    torch::Tensor coefs, bool relu) {
+  TORCH_CHECK(dz.is_cuda() && dz.dim() == 4 &&
+                  dz.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3_dgrad_bn: channels_last GPU dz");
+  TORCH_CHECK(dz.scalar_type() == torch::kBFloat16 &&
+                  w.scalar_type() == torch::kBFloat16,
+              "conv3x3_dgrad_bn: bf16 only");
+  const int N = dz.size(0), Co = dz.size(1), H = dz.size(2),
+            W = dz.size(3);
+  const int Ci = w.size(1);
+  const bool ok = (Co == Ci) && ((Co == 16 && W == 32) ||
+                                 (Co == 32 && W == 16) ||
+                                 (Co == 64 && W == 8));
+  TORCH_CHECK(ok && H % 8 == 0, "conv3x3_dgrad_bn: unsupported shape");
+  TORCH_CHECK(coefs.numel() == 3 * Co && coefs.is_cuda() &&
+                  coefs.scalar_type() == torch::kFloat,
+              "conv3x3_dgrad_bn: coefs must be fp32 [3, Co]");
+  auto dx = torch::empty(
+      {N, Ci, H, W},
+      dz.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto dyc = torch::empty_like(dz);
+  const __hip_bfloat16* dzp =
+      reinterpret_cast<const __hip_bfloat16*>(dz.data_ptr());
+  const __hip_bfloat16* wp =
+      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
+  const __hip_bfloat16* xp =
+      reinterpret_cast<const __hip_bfloat16*>(xbn.data_ptr());
+  const __hip_bfloat16* zp =
+      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr());
+  __hip_bfloat16* dxp = reinterpret_cast<__hip_bfloat16*>(dx.data_ptr());
+  __hip_bfloat16* dycp = reinterpret_cast<__hip_bfloat16*>(dyc.data_ptr());
+  const float* cp = coefs.data_ptr<float>();
+  const int rl = relu ? 1 : 0;
+  if (Co == 16)
+    hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32, true>),
+                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dzp,
+                       wp, dxp, xp, zp, cp, dycp, N, H, rl);
+  else if (Co == 32)
+    hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16, true>),
+                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dzp,
+                       wp, dxp, xp, zp, cp, dycp, N, H, rl);
+  else
+    hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8, true>),
+                       dim3(N * (H / 8) * 2), dim3(FT_BLOCK), 0, STREAM,
+                       dzp, wp, dxp, xp, zp, cp, dycp, N, H, rl);
+  return {dx, dyc};
+}
+
+
 
 // ==========================================================================
 // MFMA 3x3/s1/p1 NHWC bf16 conv weight gradient (convwrw.h)
@@ -1576,6 +1635,71 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dweight, dbias, dres};
 }
 
+// deferred BN backward: stats + per-channel affine coefs (+ optional
+// dres); the elementwise dx pass moves into the consuming conv's staging
+std::vector<torch::Tensor> bn_bwd_defer(
+    torch::Tensor dz, torch::Tensor x, torch::Tensor z,
+    torch::Tensor save_mean, torch::Tensor save_ivar, torch::Tensor weight,
+    bool relu, bool has_res) {
+  TORCH_CHECK(x.is_cuda() && dz.is_cuda() &&
+                  x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn_bwd_defer: channels_last GPU");
+  long N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto f32 = x.options().dtype(torch::kFloat);
+  auto coefs = torch::empty({3, C}, f32);
+  auto dweight = torch::empty({C}, f32);
+  auto dbias = torch::empty({C}, f32);
+  auto dres = has_res ? torch::empty_like(x)
+                      : torch::empty({0}, x.options());
+  AT_DISPATCH_FLOATING_TYPES_AND(at::ScalarType::BFloat16, x.scalar_type(),
+                                 "bnh_bwd_defer", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,
+                                 __hip_bfloat16, scalar_t>;
+    constexpr int VN = BnVec<T>::N;
+    TORCH_CHECK(bnh_ok(C, VN), "bn_bwd_defer: unsupported C=", C);
+    const int lgc = bnh_lgc(C, VN);
+    const long NI = N * HW, tasks = NI << lgc;
+    const int B = bnh_red_grid(tasks);
+    auto part = torch::empty({B, C, 2}, f32);
+    const T* yp = relu ? reinterpret_cast<const T*>(z.data_ptr()) : nullptr;
+    hipLaunchKernelGGL((bnh_bwd_stats_k<T, typename BnVec<T>::V, VN>),
+                       dim3(B), dim3(FT_BLOCK), 0, STREAM,
+                       reinterpret_cast<const T*>(dz.data_ptr()),
+                       reinterpret_cast<const T*>(x.data_ptr()), yp,
+                       nullptr, save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(), NI, C, lgc,
+                       part.data_ptr<float>(), relu ? 1 : 0);
+    torch::Tensor red = part;
+    int Bred = B;
+    if (B > 64) {
+      // collapse partials first: the 1-block coef kernel over ~1000 rows
+      // is a serial latency chain (measured 5.9 us/call, ~98 us/step)
+      const int B2 = 32;
+      red = torch::empty({B2, C, 2}, f32);
+      hipLaunchKernelGGL(part_reduce_k, dim3(B2), dim3(FT_BLOCK), 0,
+                         STREAM, part.data_ptr<float>(), B, (int)(C * 2),
+                         red.data_ptr<float>(), B2);
+      Bred = B2;
+    }
+    hipLaunchKernelGGL(bnh_bwd_coef_k, dim3(1), dim3(FT_BLOCK), 0, STREAM,
+                       red.data_ptr<float>(), Bred, NI,
+                       save_mean.data_ptr<float>(),
+                       save_ivar.data_ptr<float>(),
+                       weight.numel() ? weight.data_ptr<float>() : nullptr,
+                       C, coefs.data_ptr<float>(),
+                       dweight.data_ptr<float>(), dbias.data_ptr<float>());
+    if (has_res) {
+      const long total_v = NI * C / VN;
+      hipLaunchKernelGGL((bnh_mask_dres_k<T, typename BnVec<T>::V, VN>),
+                         dim3(ft_grid(total_v)), dim3(FT_BLOCK), 0, STREAM,
+                         reinterpret_cast<const T*>(dz.data_ptr()),
+                         reinterpret_cast<const T*>(z.data_ptr()),
+                         reinterpret_cast<T*>(dres.data_ptr()), total_v);
+    }
+  });
+  return {coefs, dweight, dbias, dres};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_fwd_train", &bn_fwd_train);
   m.def("bn_fwd_train_part", &bn_fwd_train_part);
@@ -1604,5 +1728,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_wrw", &conv3x3_wrw);
   m.def("conv3x3_wrw2", &conv3x3_wrw2);
   m.def("conv3x3_dgrad", &conv3x3_dgrad);
+  m.def("conv3x3_dgrad_bn", &conv3x3_dgrad_bn);
+  m.def("bn_bwd_defer", &bn_bwd_defer);
   m.def("conv3x3_bn_fwd", &conv3x3_bn_fwd);
 }

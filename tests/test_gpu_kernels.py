@@ -747,3 +747,54 @@ def test_conv3x3_dgrad_numerics():
             False, [0, 0], 1, [True, False, False])[0]
         rel = (dx.float() - ref).abs().max().item() / ref.abs().max().item()
         assert rel < 0.01, 'C%d rel %.5f' % (C, rel)
+
+
+def test_bn_defer_backward_matches_eager():
+    """Deferred BN backward (dx applied in the conv dgrad staging) must
+    match the eager bn_bwd + conv backward chain on a conv->BNReLU->conv
+    stack, including weight/bias grads."""
+    import os
+    import torch.nn as nn
+    from fedtorch_amd.ops.conv3x3 import NhwcConv3x3
+    from fedtorch_amd.ops import batchnorm as bnm
+    from fedtorch_amd.ops import conv3x3 as c3
+    CL = torch.channels_last
+
+    def build():
+        torch.manual_seed(21)
+        m = nn.Sequential(
+            NhwcConv3x3(16, 16, 3, padding=1, bias=False),
+            bnm.BNReLU(16),
+            NhwcConv3x3(16, 16, 3, padding=1, bias=False),
+        ).cuda().to(memory_format=CL)
+        bnm.convert_to_fused_bn(m)
+        for p_ in m.parameters():
+            if p_.dim() == 4:
+                p_.data = p_.data.bfloat16()
+        m.train()
+        return m
+
+    def run(defer):
+        old = c3._BNDEFER_ENABLED
+        c3._BNDEFER_ENABLED = defer
+        try:
+            m = build()
+            torch.manual_seed(33)
+            x = torch.randn(64, 16, 32, 32, device='cuda').to(
+                memory_format=CL).bfloat16().requires_grad_(True)
+            out = m(x)
+            loss = (out.float() ** 2).mean()
+            loss.backward()
+            grads = [x.grad.float().clone()] + \
+                [p_.grad.float().clone() for p_ in m.parameters()]
+        finally:
+            c3._BNDEFER_ENABLED = old
+        return grads
+
+    g_def = run(True)
+    g_eag = run(False)
+    assert not c3._BNBWD_TAGS, 'side table must drain'
+    for i, (a, b) in enumerate(zip(g_def, g_eag)):
+        scale = b.abs().max().item() + 1e-6
+        rel = (a - b).abs().max().item() / scale
+        assert rel < 0.02, 'grad %d rel %.5f' % (i, rel)
