@@ -39,6 +39,11 @@ _FWD_ENABLED = os.environ.get('FEDTORCH_MFMA_FWD', '1') == '1'
 # default OFF until the partial pipeline is dieted.  FEDTORCH_WRW2=1 to
 # enable.
 _WRW2_ENABLED = os.environ.get('FEDTORCH_WRW2', '0') == '1'
+# per-shape wrw2: the v2 kernel beats MIOpen-incl-wrappers at C16
+# (23.4 vs 33 us) but not at C32/C64 — use it where it wins
+_WRW2_SHAPES = set(
+    int(c) for c in os.environ.get('FEDTORCH_WRW2_C', '16').split(',')
+    if c.strip())
 # MFMA direct conv backward-data (hip/convfwd.h conv3x3_dgrad_k): the fwd
 # kernel's mirror, 5.0-6.7 us/call vs MIOpen's 23-24.  Default ON.
 _DGRAD_ENABLED = os.environ.get('FEDTORCH_MFMA_DGRAD', '1') == '1'
@@ -108,12 +113,13 @@ class _Conv3x3BNFn(torch.autograd.Function):
                     dyc, x, weight, None, [1, 1], [1, 1], [1, 1], False,
                     [0, 0], 1, [False, True, False])[1]
             return dx, dw
+        use_wrw2 = _WRW2_ENABLED or weight.shape[0] in _WRW2_SHAPES
         custom_dx = _DGRAD_ENABLED
-        custom_dw = _WRW2_ENABLED or _ENABLED
+        custom_dw = use_wrw2 or _ENABLED
         if custom_dx and custom_dw:
             dx = ops._C.conv3x3_dgrad(dy, weight) \
                 if ctx.needs_input_grad[0] else None
-            dw = ops._C.conv3x3_wrw2(dy, x) if _WRW2_ENABLED \
+            dw = ops._C.conv3x3_wrw2(dy, x) if use_wrw2 \
                 else ops._C.conv3x3_wrw(dy, x)
         elif custom_dx:
             dx = ops._C.conv3x3_dgrad(dy, weight) \
