@@ -144,6 +144,32 @@ class _Conv3x3BNFn(torch.autograd.Function):
         return dx, dw
 
 
+class _Conv3x3S2BNFn(torch.autograd.Function):
+    """stride-2 transition conv (16->32, 32->64) with the fused BN-stats
+    epilogue; backward = ONE combined MIOpen call (these run once per
+    step each — the fwd CK solver + the following BN's stats pass are
+    the pool)."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        H_out = x.shape[2] // 2
+        grid = x.shape[0] * (H_out // 8) * (weight.shape[0] // 32)
+        part = torch.empty(grid, weight.shape[0], 2, device=x.device)
+        y = ops._C.conv3x3s2_bn_fwd(x, weight, part)
+        ctx.save_for_backward(x, weight)
+        ctx.mark_non_differentiable(part)
+        return y, part
+
+    @staticmethod
+    def backward(ctx, dy, _dpart):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=_CL)
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, weight, None, [2, 2], [1, 1], [1, 1], False, [0, 0],
+            1, [bool(ctx.needs_input_grad[0]), True, False])
+        return dx, dw
+
+
 class _Conv3x3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight):
@@ -185,6 +211,22 @@ class NhwcConv3x3(nn.Conv2d):
             if wb.is_contiguous(memory_format=_CL):
                 y, part = _Conv3x3BNFn.apply(x, wb)
                 y._ft_bn_part = part  # consumed by FusedBatchNorm2d
+                return y
+        use_s2 = (_FWD_ENABLED and self.training and x.is_cuda
+                  and x.dim() == 4 and self.bias is None
+                  and self.stride == (2, 2) and self.padding == (1, 1)
+                  and x.dtype == torch.bfloat16
+                  and self.out_channels == 2 * self.in_channels
+                  and (self.in_channels, x.shape[3]) in ((16, 32), (32, 16))
+                  and x.shape[2] % 16 == 0
+                  and x.is_contiguous(memory_format=_CL)
+                  and ops.hip_available() and not ops.FORCE_EAGER)
+        if use_s2:
+            wb = w if w.dtype == torch.bfloat16 else \
+                w.bfloat16().contiguous(memory_format=_CL)
+            if wb.is_contiguous(memory_format=_CL):
+                y, part = _Conv3x3S2BNFn.apply(x, wb)
+                y._ft_bn_part = part
                 return y
         use = (_ENABLED and torch.is_grad_enabled() and self.training
                and x.is_cuda

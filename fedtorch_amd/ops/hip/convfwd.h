@@ -31,22 +31,24 @@
 
 // bf16x8 / f32x4 come from convwrw.h (same TU)
 
-template <int CI, int CO, int CO_TILE, int W, bool FUSE_IN, bool HAS_RES>
+template <int CI, int CO, int CO_TILE, int W, bool FUSE_IN, bool HAS_RES,
+          int S = 1>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
-    const __hip_bfloat16* __restrict__ x,   // [N, H, W, CI]
+    const __hip_bfloat16* __restrict__ x,   // [N, H*S, W*S, CI]
     const __hip_bfloat16* __restrict__ w,   // [CO, 3, 3, CI]
-    __hip_bfloat16* __restrict__ y,         // [N, H, W, CO]
+    __hip_bfloat16* __restrict__ y,         // [N, H, W, CO]  (H, W = OUT)
     float* __restrict__ ysum,               // [2*CO] (sum, sumsq) or null
     const float* __restrict__ in_a,         // [CI] scale or null
     const float* __restrict__ in_b,         // [CI] shift
-    const __hip_bfloat16* __restrict__ res, // [N, H, W, CI] residual or null
+    const __hip_bfloat16* __restrict__ res, // residual or null
     int N, int H, int relu_in) {
   constexpr int R = 8;                    // output rows per workgroup
+  constexpr int WI = W * S;               // input row width
   constexpr int KTOT = 9 * CI;
   constexpr int KPAD = (KTOT + 31) & ~31; // MFMA K granularity
   constexpr int K8 = KPAD / 8;
   constexpr int CIP = CI + 8;             // padded pixel slot (elements)
-  constexpr int XR = R + 2, XC = W + 2;
+  constexpr int XR = R * S + 2, XC = WI + 2;
   constexpr int NSLOT = XR * XC + 1;      // +1 zero slot for K padding
   constexpr int P = R * W;                // output pixels per WG
   constexpr int MT = P / 16;              // M-tiles
@@ -67,7 +69,8 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
   const int rb = (blockIdx.x / (CO / CO_TILE)) % (H / R);
   const int n = blockIdx.x / ((CO / CO_TILE) * (H / R));
   const int co0 = cob * CO_TILE;
-  const int h0 = rb * R;
+  const int h0 = rb * R;                  // OUTPUT row base
+  const int HI = H * S;                   // input height
 
   if (FUSE_IN && tid < CI) {
     sa[tid] = in_a[tid];
@@ -106,16 +109,16 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
   }
   __syncthreads();  // zero-fill visible before body writes land below
 
-  // body: XR rows x W*CI contiguous elements each (16 B chunks)
-  constexpr int BCH = W * CI / 8;
+  // body: XR input rows x WI*CI contiguous elements each (16 B chunks)
+  constexpr int BCH = WI * CI / 8;
   const float zero = 0.f;
   for (int e = tid; e < XR * BCH; e += FT_BLOCK) {
     const int row = e / BCH, c = e % BCH;
-    const int hh = h0 - 1 + row;
+    const int hh = h0 * S - 1 + row;
     uint4 v = {0, 0, 0, 0};
-    if (hh >= 0 && hh < H)
+    if (hh >= 0 && hh < HI)
       v = *reinterpret_cast<const uint4*>(
-          x + (((long)n * H + hh) * W) * CI + c * 8);
+          x + (((long)n * HI + hh) * WI) * CI + c * 8);
     const int pix = c * 8 / CI;          // pixel within the row
     const int ci0 = c * 8 % CI;
     if (FUSE_IN) {
@@ -125,11 +128,11 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
       // HAS_RES are COMPILE-TIME: a runtime branch here makes hipcc
       // branch around each staging load and drain vmcnt per element
       // (guide §5 trap (c); measured +12 us per call).
-      if (hh >= 0 && hh < H) {
+      if (hh >= 0 && hh < HI) {
         uint4 rv = {0, 0, 0, 0};
         if (HAS_RES)
           rv = *reinterpret_cast<const uint4*>(
-              res + (((long)n * H + hh) * W) * CI + c * 8);
+              res + (((long)n * HI + hh) * WI) * CI + c * 8);
         const __hip_bfloat16* xv =
             reinterpret_cast<const __hip_bfloat16*>(&v);
         const __hip_bfloat16* rr =
@@ -167,7 +170,8 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
     for (int mt = 0; mt < MTW; ++mt) {
       const int p = (wave * MTW + mt) * 16 + fm;
       const int r = p / W, c = p % W;
-      const int slot = (tap < 9) ? ((r + dh) * XC + (c + dw)) : (NSLOT - 1);
+      const int slot = (tap < 9) ? ((r * S + dh) * XC + (c * S + dw))
+                                 : (NSLOT - 1);
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
           &xs[(long)slot * CIP + ci0]);
 #pragma unroll

@@ -1026,6 +1026,51 @@ torch::Tensor conv3x3_bn_fwd(torch::Tensor x, torch::Tensor w,
 }
 
 // ==========================================================================
+// MFMA fused conv fwd, STRIDE 2 (the ResNet transition convs)
+// ==========================================================================
+torch::Tensor conv3x3s2_bn_fwd(torch::Tensor x, torch::Tensor w,
+                               torch::Tensor ysum) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 4 &&
+                  x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "conv3x3s2_bn_fwd: channels_last GPU");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+                  w.scalar_type() == torch::kBFloat16,
+              "conv3x3s2_bn_fwd: bf16 only");
+  const int N = x.size(0), Ci = x.size(1), HI = x.size(2), WI = x.size(3);
+  const int Co = w.size(0);
+  const int H = HI / 2, W = WI / 2;
+  const bool ok = (Co == 2 * Ci) && ((Ci == 16 && WI == 32) ||
+                                     (Ci == 32 && WI == 16));
+  TORCH_CHECK(ok, "conv3x3s2_bn_fwd: unsupported (Ci,Co,WI)=", Ci, ",",
+              Co, ",", WI);
+  TORCH_CHECK(H % 8 == 0, "conv3x3s2_bn_fwd: H_out % 8 != 0");
+  auto y = torch::empty(
+      {N, Co, H, W},
+      x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  float* ysp = ysum.defined() && ysum.numel() > 0
+                   ? ysum.data_ptr<float>() : nullptr;
+  const int grid = N * (H / 8) * (Co / 32);
+  if (ysp)
+    TORCH_CHECK(ysum.numel() == (long)grid * Co * 2,
+                "conv3x3s2_bn_fwd: ysum must be [grid, Co, 2], grid=",
+                grid);
+  const __hip_bfloat16* xp =
+      reinterpret_cast<const __hip_bfloat16*>(x.data_ptr());
+  const __hip_bfloat16* wp =
+      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr());
+  __hip_bfloat16* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
+  if (Ci == 16)
+    hipLaunchKernelGGL((conv3x3_bn_fwd_k<16, 32, 32, 16, false, false, 2>),
+                       dim3(grid), dim3(FT_BLOCK), 0, STREAM, xp, wp, yp,
+                       ysp, nullptr, nullptr, nullptr, N, H, 0);
+  else
+    hipLaunchKernelGGL((conv3x3_bn_fwd_k<32, 64, 32, 8, false, false, 2>),
+                       dim3(grid), dim3(FT_BLOCK), 0, STREAM, xp, wp, yp,
+                       ysp, nullptr, nullptr, nullptr, N, H, 0);
+  return y;
+}
+
+// ==========================================================================
 // MFMA direct conv backward-data (convfwd.h conv3x3_dgrad_k)
 // ==========================================================================
 torch::Tensor conv3x3_dgrad(torch::Tensor dy, torch::Tensor w) {
@@ -1731,4 +1776,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_dgrad_bn", &conv3x3_dgrad_bn);
   m.def("bn_bwd_defer", &bn_bwd_defer);
   m.def("conv3x3_bn_fwd", &conv3x3_bn_fwd);
+  m.def("conv3x3s2_bn_fwd", &conv3x3s2_bn_fwd);
 }

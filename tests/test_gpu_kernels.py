@@ -798,3 +798,26 @@ def test_bn_defer_backward_matches_eager():
         scale = b.abs().max().item() + 1e-6
         rel = (a - b).abs().max().item() / scale
         assert rel < 0.02, 'grad %d rel %.5f' % (i, rel)
+
+
+def test_conv3x3_s2_fused_fwd():
+    """stride-2 transition conv (fused BN-stats epilogue) vs fp32 ref."""
+    import torch.nn.functional as F
+    CL = torch.channels_last
+    torch.manual_seed(15)
+    for Ci, Wi in ((16, 32), (32, 16)):
+        Co, N = 2 * Ci, 64
+        x = torch.randn(N, Ci, Wi, Wi, device='cuda').to(
+            memory_format=CL).bfloat16()
+        w = (torch.randn(Co, Ci, 3, 3, device='cuda') / (3 * Ci) ** .5).to(
+            memory_format=CL).bfloat16()
+        grid = N * (Wi // 2 // 8) * (Co // 32)
+        part = torch.zeros(grid, Co, 2, device='cuda')
+        y = ops._C.conv3x3s2_bn_fwd(x, w, part)
+        ref = F.conv2d(x.float(), w.float(), None, 2, 1)
+        rel = (y.float() - ref).abs().max().item() / ref.abs().max().item()
+        assert rel < 0.01, 'Ci%d rel %.5f' % (Ci, rel)
+        s_ref = y.float().sum(dim=(0, 2, 3))
+        e = (part[:, :, 0].sum(0) - s_ref).abs().max().item() / (
+            (y.float() ** 2).sum().item() ** 0.5 + 1e-6)
+        assert e < 1e-2, 'stats %.5f' % e
