@@ -98,12 +98,18 @@ class _FusedBNFunction(torch.autograd.Function):
             # transformed dy for its wrw) — return dz tagged via the
             # side table instead of running bnh_bwd_dx here
             z = y if ctx.relu else None
-            coefs, dweight, dbias, dres = ops._C.bn_bwd_defer(
+            # dres is ALLOCATED here but FILLED by the consuming conv's
+            # staging kernel (stream order guarantees the fill lands
+            # before any accumulation reads it); bn_bwd_defer only runs
+            # its mask kernel when the conv cannot fill (defensive: the
+            # registry consumer always can on this path)
+            coefs, dweight, dbias, _ = ops._C.bn_bwd_defer(
                 dy, x, z if z is not None else empty,
                 save_mean, save_ivar,
                 weight if weight is not None else empty,
-                ctx.relu, ctx.has_res)
-            _c3.register_bn_defer(dy, x, z, coefs, ctx.relu)
+                ctx.relu, False)
+            dres = torch.empty_like(dy) if ctx.has_res else None
+            _c3.register_bn_defer(dy, x, z, coefs, ctx.relu, dres)
             return (dy, dweight if weight is not None else None,
                     dbias if weight is not None else None, None, None,
                     None, None, None,

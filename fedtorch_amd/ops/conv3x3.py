@@ -66,8 +66,8 @@ def bn_defer_active():
     return _DGRAD_ENABLED and _BNDEFER_ENABLED and not ops.FORCE_EAGER
 
 
-def register_bn_defer(grad, xbn, z, coefs, relu):
-    _BNBWD_TAGS[grad.data_ptr()] = (xbn, z, coefs, relu)
+def register_bn_defer(grad, xbn, z, coefs, relu, dres=None):
+    _BNBWD_TAGS[grad.data_ptr()] = (xbn, z, coefs, relu, dres)
 _EMPTY = {}
 
 
@@ -101,9 +101,11 @@ class _Conv3x3BNFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=_CL)
         tag = _BNBWD_TAGS.pop(dy.data_ptr(), None) if _BNBWD_TAGS else None
         if tag is not None and _DGRAD_ENABLED:
-            xbn, z, coefs, relu = tag
+            xbn, z, coefs, relu, dres = tag
+            e = _empty(dy.device)
             dx, dyc = ops._C.conv3x3_dgrad_bn(
-                dy, weight, xbn, z if z is not None else xbn, coefs, relu)
+                dy, weight, xbn, z if z is not None else xbn, coefs, relu,
+                dres if dres is not None else e)
             if not ctx.needs_input_grad[0]:
                 dx = None
             if _WRW2_ENABLED:
@@ -164,6 +166,23 @@ class _Conv3x3S2BNFn(torch.autograd.Function):
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=_CL)
+        tag = _BNBWD_TAGS.pop(dy.data_ptr(), None) if _BNBWD_TAGS else None
+        if tag is not None:
+            # a deferred BN backward landed on the stride-2 conv (no TRF
+            # kernel here): materialize dy_conv = A*mask(dz) + B + D*x_bn
+            # eagerly (2 transition convs per step) and fill dres
+            xbn, z, coefs, relu, dres = tag
+            C = dy.shape[1]
+            g = dy
+            if relu and z is not None:
+                g = dy * (z > 0)
+            if dres is not None:
+                dres.copy_(g)
+            A = coefs[0].view(1, C, 1, 1)
+            B = coefs[1].view(1, C, 1, 1)
+            D = coefs[2].view(1, C, 1, 1)
+            dy = (g.float() * A + B + D * xbn.float()).bfloat16() \
+                .contiguous(memory_format=_CL)
         dx, dw, _ = torch.ops.aten.convolution_backward(
             dy, x, weight, None, [2, 2], [1, 1], [1, 1], False, [0, 0],
             1, [bool(ctx.needs_input_grad[0]), True, False])

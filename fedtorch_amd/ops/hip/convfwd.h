@@ -254,7 +254,8 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_bn_fwd_k(
 // gathered as [k/8][CI_TILE][8] with the tap REVERSED (full correlation
 // <-> convolution flip), K = 9*CO resident in LDS.
 // ==========================================================================
-template <int CI, int CO, int CI_TILE, int W, bool TRF>
+template <int CI, int CO, int CI_TILE, int W, bool TRF,
+          bool HAS_DRES = false>
 __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
     const __hip_bfloat16* __restrict__ dy,  // [N, H, W, CO] (dz when TRF)
     const __hip_bfloat16* __restrict__ w,   // [CO, 3, 3, CI]
@@ -263,6 +264,7 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
     const __hip_bfloat16* __restrict__ zv,  // BN output (ReLU mask src)
     const float* __restrict__ coefs,        // [3, CO] A, B, D
     __hip_bfloat16* __restrict__ dyc,       // transformed dy out (for wrw)
+    __hip_bfloat16* __restrict__ dres,      // masked dz out (residual)
     int N, int H, int relu) {
   constexpr int R = 8;
   constexpr int KTOT = 9 * CO;
@@ -352,15 +354,19 @@ __global__ void __launch_bounds__(FT_BLOCK) conv3x3_dgrad_k(
           reinterpret_cast<const __hip_bfloat16*>(&xv4);
       const __hip_bfloat16* zz =
           reinterpret_cast<const __hip_bfloat16*>(&zv4);
-      __hip_bfloat16 out[8];
+      __hip_bfloat16 out[8], gm[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)gg[j];
         if (relu && (float)zz[j] <= 0.f) g = 0.f;
+        if (HAS_DRES) gm[j] = (__hip_bfloat16)g;
         out[j] = (__hip_bfloat16)(
             fmaf(sA[co8 + j], g,
                  fmaf(sD[co8 + j], (float)xx[j], sB[co8 + j])));
       }
+      if (HAS_DRES)
+        *reinterpret_cast<uint4*>(dres + gbase) =
+            *reinterpret_cast<const uint4*>(gm);
       v = *reinterpret_cast<const uint4*>(out);
       *reinterpret_cast<uint4*>(dyc + gbase) = v;
     }

@@ -1101,18 +1101,18 @@ torch::Tensor conv3x3_dgrad(torch::Tensor dy, torch::Tensor w) {
   if (Co == 16)
     hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32, false>),
                        dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
-                       wp, dxp, nullptr, nullptr, nullptr, nullptr, N, H,
-                       0);
+                       wp, dxp, nullptr, nullptr, nullptr, nullptr,
+                       nullptr, N, H, 0);
   else if (Co == 32)
     hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16, false>),
                        dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dyp,
-                       wp, dxp, nullptr, nullptr, nullptr, nullptr, N, H,
-                       0);
+                       wp, dxp, nullptr, nullptr, nullptr, nullptr,
+                       nullptr, N, H, 0);
   else
     hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8, false>),
                        dim3(N * (H / 8) * 2), dim3(FT_BLOCK), 0, STREAM,
                        dyp, wp, dxp, nullptr, nullptr, nullptr, nullptr,
-                       N, H, 0);
+                       nullptr, N, H, 0);
   return dx;
 }
 
@@ -1122,7 +1122,7 @@ torch::Tensor conv3x3_dgrad(torch::Tensor dy, torch::Tensor w) {
 // bnh_bwd_dx pass disappears.
 std::vector<torch::Tensor> conv3x3_dgrad_bn(
     torch::Tensor dz, torch::Tensor w, torch::Tensor xbn, torch::Tensor z,
-    torch::Tensor coefs, bool relu) {
+    torch::Tensor coefs, bool relu, torch::Tensor dres) {
   TORCH_CHECK(dz.is_cuda() && dz.dim() == 4 &&
                   dz.is_contiguous(at::MemoryFormat::ChannelsLast),
               "conv3x3_dgrad_bn: channels_last GPU dz");
@@ -1143,6 +1143,11 @@ std::vector<torch::Tensor> conv3x3_dgrad_bn(
       {N, Ci, H, W},
       dz.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto dyc = torch::empty_like(dz);
+  const bool want_dres = dres.defined() && dres.numel() > 0;
+  if (want_dres)
+    TORCH_CHECK(dres.sizes() == dz.sizes() &&
+                    dres.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "conv3x3_dgrad_bn: dres must match dz, channels_last");
   const __hip_bfloat16* dzp =
       reinterpret_cast<const __hip_bfloat16*>(dz.data_ptr());
   const __hip_bfloat16* wp =
@@ -1153,20 +1158,33 @@ std::vector<torch::Tensor> conv3x3_dgrad_bn(
       reinterpret_cast<const __hip_bfloat16*>(z.data_ptr());
   __hip_bfloat16* dxp = reinterpret_cast<__hip_bfloat16*>(dx.data_ptr());
   __hip_bfloat16* dycp = reinterpret_cast<__hip_bfloat16*>(dyc.data_ptr());
+  __hip_bfloat16* drp = want_dres
+      ? reinterpret_cast<__hip_bfloat16*>(dres.data_ptr()) : nullptr;
   const float* cp = coefs.data_ptr<float>();
   const int rl = relu ? 1 : 0;
-  if (Co == 16)
-    hipLaunchKernelGGL((conv3x3_dgrad_k<16, 16, 16, 32, true>),
-                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dzp,
-                       wp, dxp, xp, zp, cp, dycp, N, H, rl);
-  else if (Co == 32)
-    hipLaunchKernelGGL((conv3x3_dgrad_k<32, 32, 32, 16, true>),
-                       dim3(N * (H / 8)), dim3(FT_BLOCK), 0, STREAM, dzp,
-                       wp, dxp, xp, zp, cp, dycp, N, H, rl);
-  else
-    hipLaunchKernelGGL((conv3x3_dgrad_k<64, 64, 32, 8, true>),
-                       dim3(N * (H / 8) * 2), dim3(FT_BLOCK), 0, STREAM,
-                       dzp, wp, dxp, xp, zp, cp, dycp, N, H, rl);
+#define FT_DG_LAUNCH(CI_, CO_, CT_, W_, XN_)                              \
+  {                                                                       \
+    if (want_dres)                                                        \
+      hipLaunchKernelGGL((conv3x3_dgrad_k<CI_, CO_, CT_, W_, true,        \
+                                          true>),                        \
+                         dim3(N * (H / 8) * XN_), dim3(FT_BLOCK), 0,      \
+                         STREAM, dzp, wp, dxp, xp, zp, cp, dycp, drp, N,  \
+                         H, rl);                                          \
+    else                                                                  \
+      hipLaunchKernelGGL((conv3x3_dgrad_k<CI_, CO_, CT_, W_, true,        \
+                                          false>),                       \
+                         dim3(N * (H / 8) * XN_), dim3(FT_BLOCK), 0,      \
+                         STREAM, dzp, wp, dxp, xp, zp, cp, dycp, drp, N,  \
+                         H, rl);                                          \
+  }
+  if (Co == 16) {
+    FT_DG_LAUNCH(16, 16, 16, 32, 1)
+  } else if (Co == 32) {
+    FT_DG_LAUNCH(32, 32, 32, 16, 1)
+  } else {
+    FT_DG_LAUNCH(64, 64, 32, 8, 2)
+  }
+#undef FT_DG_LAUNCH
   return {dx, dyc};
 }
 

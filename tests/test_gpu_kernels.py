@@ -821,3 +821,47 @@ def test_conv3x3_s2_fused_fwd():
         e = (part[:, :, 0].sum(0) - s_ref).abs().max().item() / (
             (y.float() ** 2).sum().item() ** 0.5 + 1e-6)
         assert e < 1e-2, 'stats %.5f' % e
+
+
+def test_bn_defer_full_resnet_matches_eager():
+    """Deferred BN backward across a FULL ResNet-20 step (incl. stride-2
+    transitions, residual forks and BNAddReLU dres) vs the eager path."""
+    from types import SimpleNamespace
+    from fedtorch_amd.ops import conv3x3 as c3
+    from fedtorch_amd.components.models.resnet import resnet
+    from fedtorch_amd.ops.batchnorm import convert_to_fused_bn
+    CL = torch.channels_last
+
+    def run(defer):
+        old = c3._BNDEFER_ENABLED
+        c3._BNDEFER_ENABLED = defer
+        try:
+            torch.manual_seed(41)
+            args = SimpleNamespace(arch='resnet20', data='cifar10')
+            m = resnet(args).cuda().to(memory_format=CL)
+            convert_to_fused_bn(m)
+            for p_ in m.parameters():
+                if p_.dim() == 4:
+                    p_.data = p_.data.bfloat16()
+            m.train()
+            torch.manual_seed(42)
+            x = torch.randn(32, 3, 32, 32, device='cuda').to(
+                memory_format=CL)
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                out = m(x)
+                loss = out.float().square().mean()
+            loss.backward()
+            gs = [p_.grad.float().clone() for p_ in m.parameters()
+                  if p_.grad is not None]
+        finally:
+            c3._BNDEFER_ENABLED = old
+        return gs
+
+    g_d = run(True)
+    g_e = run(False)
+    assert not c3._BNBWD_TAGS, 'side table must drain'
+    assert len(g_d) == len(g_e)
+    for i, (a, b) in enumerate(zip(g_d, g_e)):
+        scale = b.abs().max().item() + 1e-5
+        rel = (a - b).abs().max().item() / scale
+        assert rel < 0.05, 'param %d rel %.4f' % (i, rel)
