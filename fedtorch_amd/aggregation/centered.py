@@ -39,13 +39,18 @@ def fedavg_aggregation_centered(Clients, Server, online_clients,
     for o in online_clients:
         diff = Server.work.setdefault('diff', Server.arena.new_buffer())
         ops.scaled_diff(Server.arena.flat, Clients[o].arena.flat, diff, w[o])
-        if args.federated_type == 'fedadam':
-            from fedtorch_amd.aggregation.federated import _fedadam_normalize
-            _fedadam_normalize(Clients[o].args, Server.arena, diff)
         if args.quantized:
             q, info = ops.quantize(diff, args.quantized_bits)
             diff = ops.dequantize(q, info)
         agg.add_(diff)
+    if args.federated_type == 'fedadam':
+        # normalize the AGGREGATE with the server-side v state — the
+        # reference applies fedadam on the summed diff at the server
+        # (`federated/fedavg.py:81-85`); normalizing per-client diffs
+        # (the r1 behavior) made the two execution modes disagree
+        # (caught by tests/test_equivalence.py)
+        from fedtorch_amd.aggregation.federated import _fedadam_normalize
+        _fedadam_normalize(Server.args, Server.arena, agg)
     Server.optimizer.step(apply_lr=False, scale=args.lr_scale_at_sync,
                           apply_in_momentum=False,
                           apply_out_momentum=args.out_momentum, grad=agg)

@@ -32,6 +32,8 @@ def _argv(fed_type, j, ckpt):
                                 '--checkpoint', ckpt]
     if fed_type == 'fedgate':
         argv += ['--compressed', 'false']
+    if fed_type == 'fedadam':
+        argv += ['--fedadam_beta', '0.9', '--fedadam_tau', '0.1']
     return argv
 
 
@@ -92,7 +94,9 @@ def _dist_worker(rank, world, port, fed_type, q):
 
 @pytest.mark.parametrize('fed_type,port', [('fedavg', 29931),
                                            ('scaffold', 29933),
-                                           ('fedgate', 29935)])
+                                           ('fedgate', 29935),
+                                           ('qsparse', 29937),
+                                           ('fedadam', 29939)])
 def test_centered_equals_distributed(fed_type, port):
     ctx = mp.get_context('spawn')
 
