@@ -77,9 +77,11 @@ class _ResNetBase(nn.Module):
     def _make_stage(self, block_fn, planes, block_num, stride=1):
         downsample = None
         if stride != 1 or self.inplanes != planes * block_fn.expansion:
+            from fedtorch_amd.ops.conv3x3 import NhwcConv1x1S2
+            conv_cls = NhwcConv1x1S2 if stride == 2 else nn.Conv2d
             downsample = nn.Sequential(
-                nn.Conv2d(self.inplanes, planes * block_fn.expansion,
-                          kernel_size=1, stride=stride, bias=False),
+                conv_cls(self.inplanes, planes * block_fn.expansion,
+                         kernel_size=1, stride=stride, bias=False),
                 nn.BatchNorm2d(planes * block_fn.expansion))
         layers = [block_fn(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block_fn.expansion

@@ -213,6 +213,72 @@ class _Conv3x3Fn(torch.autograd.Function):
         return dx, dw
 
 
+class _Conv1x1S2Fn(torch.autograd.Function):
+    """1x1 stride-2 downsample conv as three library GEMMs over the
+    even-subsampled input (the ResNet transition shortcut): MIOpen's CK
+    solvers for this shape carry batched-GEMM + wrapper kernels; a plain
+    hipBLASLt GEMM over the strided slice is smaller and simpler.
+    y[p, co] = sum_ci x_even[p, ci] w[co, ci]."""
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        N, Ci, H, W = x.shape
+        Co = weight.shape[0]
+        xe = x[:, :, ::2, ::2].contiguous(memory_format=_CL)
+        wm = weight.reshape(Co, Ci)
+        # channels_last [N,C,H,W] -> NHWC rows: permute view, flatten
+        xr = xe.permute(0, 2, 3, 1).reshape(-1, Ci)
+        y = (xr @ wm.t()).reshape(N, H // 2, W // 2, Co) \
+            .permute(0, 3, 1, 2).contiguous(memory_format=_CL)
+        ctx.save_for_backward(xe, weight)
+        ctx.in_shape = (N, Ci, H, W)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xe, weight = ctx.saved_tensors
+        N, Ci, H, W = ctx.in_shape
+        Co = weight.shape[0]
+        dy = dy.contiguous(memory_format=_CL)
+        dyr = dy.permute(0, 2, 3, 1).reshape(-1, Co)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dxe = (dyr @ weight.reshape(Co, Ci)) \
+                .reshape(N, H // 2, W // 2, Ci).permute(0, 3, 1, 2)
+            dx = torch.zeros(N, Ci, H, W, dtype=dy.dtype, device=dy.device
+                             ).contiguous(memory_format=_CL)
+            dx[:, :, ::2, ::2] = dxe
+        xr = xe.permute(0, 2, 3, 1).reshape(-1, Ci)
+        dw = (dyr.t() @ xr).reshape(Co, Ci, 1, 1) \
+            .contiguous(memory_format=_CL)
+        return dx, dw
+
+
+_CONV1X1_GEMM = os.environ.get('FEDTORCH_CONV1X1_GEMM', '0') == '1'
+
+
+class NhwcConv1x1S2(nn.Conv2d):
+    """Drop-in 1x1/stride-2 bias-free Conv2d (the ResNet downsample
+    shortcut).  The GEMM path (default OFF) measured SLOWER in-bench
+    (160.0k vs 177.0k flagship): the K=16384 reduction wrw GEMM and the
+    strided dx scatter cost more than MIOpen's CK solvers here."""
+
+    def forward(self, x):
+        w = self.weight
+        use = (_CONV1X1_GEMM
+               and self.training and x.is_cuda and x.dim() == 4
+               and self.bias is None and self.stride == (2, 2)
+               and self.kernel_size == (1, 1)
+               and x.dtype == torch.bfloat16
+               and x.is_contiguous(memory_format=_CL)
+               and not ops.FORCE_EAGER)
+        if use:
+            wb = w if w.dtype == torch.bfloat16 else w.bfloat16()
+            return _Conv1x1S2Fn.apply(x, wb)
+        return F.conv2d(x, w, self.bias, self.stride, self.padding,
+                        self.dilation, self.groups)
+
+
 class NhwcConv3x3(nn.Conv2d):
     """Drop-in 3x3/s1/p1 bias-free Conv2d whose channels_last bf16 GPU path
     uses the MFMA wrw kernel.  state_dict layout is the stock Conv2d's."""

@@ -865,3 +865,36 @@ def test_bn_defer_full_resnet_matches_eager():
         scale = b.abs().max().item() + 1e-5
         rel = (a - b).abs().max().item() / scale
         assert rel < 0.05, 'param %d rel %.4f' % (i, rel)
+
+
+def test_conv1x1_s2_gemm_path():
+    """1x1/stride-2 downsample on the GEMM path: fwd + both grads vs
+    the fp32 conv reference."""
+    import torch.nn.functional as F
+    from fedtorch_amd.ops import conv3x3 as c3
+    from fedtorch_amd.ops.conv3x3 import NhwcConv1x1S2
+    CL = torch.channels_last
+    old = c3._CONV1X1_GEMM
+    c3._CONV1X1_GEMM = True
+    torch.manual_seed(17)
+    for Ci, Wi in ((16, 32), (32, 16)):
+        Co = 2 * Ci
+        m = NhwcConv1x1S2(Ci, Co, kernel_size=1, stride=2,
+                          bias=False).cuda().to(memory_format=CL)
+        m.weight.data = m.weight.data.bfloat16()
+        m.train()
+        x = torch.randn(64, Ci, Wi, Wi, device='cuda').to(
+            memory_format=CL).bfloat16().requires_grad_(True)
+        y = m(x)
+        loss = (y.float() ** 2).mean()
+        loss.backward()
+        xr = x.detach().float().requires_grad_(True)
+        w32 = m.weight.detach().float().requires_grad_(True)
+        yr = F.conv2d(xr, w32, None, 2, 0)
+        (yr ** 2).mean().backward()
+        assert (y.float() - yr).abs().max().item() < 0.05
+        assert (x.grad.float() - xr.grad).abs().max().item() < 0.02
+        rel = (m.weight.grad.float() - w32.grad).abs().max().item() / \
+            w32.grad.abs().max().item()
+        assert rel < 0.02, 'dw rel %.5f' % rel
+    c3._CONV1X1_GEMM = old
