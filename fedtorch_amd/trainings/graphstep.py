@@ -67,14 +67,13 @@ class GraphStepper(object):
         if self.g1 is not None:
             if _input.shape != self.sx.shape or \
                     _input.dtype != self.sx.dtype:
-                # odd batch: eager step, keep graphs — the eager path
-                # accumulates into the arena grads, so re-attach them
-                self._restore_eager()
-                return False
+                # odd batch: keep graphs, take the step on the eager
+                # STOLEN-GATHER flow (a plain eager step would need
+                # fp32 p.grad views, which the bf16 twins forbid)
+                return self._eager_stolen(_input, _target)
             if lr != self.lr0:
                 self._invalidate()      # LR moved: re-capture at new LR
-                self._restore_eager()
-                return False
+                return self._eager_stolen(_input, _target)
         if self.g1 is None:
             if self.recaptures >= self.MAX_RECAPTURE:
                 self.fail = True
@@ -115,7 +114,40 @@ class GraphStepper(object):
     def _invalidate(self):
         self.g1 = self.g2 = None
 
+    def _metric_acc(self, out, loss, y, bs):
+        with torch.no_grad():
+            od = out.detach().float()
+            _, pred = od.topk(max(self.topk), 1, True, True)
+            eq = pred.t().eq(y.view(1, -1))
+            self.macc[0] += loss.detach().float() * bs
+            for i, k in enumerate(self.topk):
+                self.macc[1 + i] += eq[:k].sum().float()
+            self.macc[-1] += bs
+
+    def _eager_stolen(self, x, y):
+        """One eager local step on the stolen-gather flow (twin-safe);
+        used for odd batches / LR-transition steps while graphs exist."""
+        c = self.client
+        args = self.args
+        c.arena.detach_grads()
+        with amp(args):
+            out = c.model(x)
+            loss = c.criterion(out, y)
+        loss.backward()
+        c.arena.gather_grads()
+        c.optimizer.step(apply_lr=True,
+                         apply_in_momentum=args.in_momentum,
+                         apply_out_momentum=False)
+        if self.macc is not None:
+            self._metric_acc(out, loss, y, float(x.size(0)))
+            self.pending += 1
+        return True
+
     def _restore_eager(self):
+        """Permanent-fail path: hand the client back to the classic
+        eager loop (fp32 params + attached fp32 grad views)."""
+        if self.client.arena.half_flat is not None:
+            self.client.arena.disable_bf16_compute()
         if self.client.arena.grad is not None:
             self.client.arena.attach_grads()
 
@@ -204,8 +236,274 @@ class GraphStepper(object):
             c.optimizer.step(apply_lr=True,
                              apply_in_momentum=args.in_momentum,
                              apply_out_momentum=False)
+        # PIN the capture-era chunk-table tensors: a later detach_grads
+        # (eager fallback) drops arena._gather_state and the captured
+        # kernel would read freed-and-reused memory at replay
+        self._keep = [arena._gather_state]
         torch.cuda.synchronize()
         self._restore(snap)
         self.macc.zero_()
         self.g1, self.g2 = g1, g2
+        self.recaptures += 1
+
+
+class GraphStepperAPFL(object):
+    """hipGraph-captured APFL local step (two models, two optimizers).
+
+    Four graphs replayed per local step: {global fwd+bwd}, {gather+step},
+    {personal BLEND fwd+bwd}, {gather+step}.  The personal forward reads
+    the global arena the in-graph step just updated (same buffers), and
+    the blend factor alpha lives in a DEVICE tensor so the per-round
+    adaptive-alpha update (`apfl_alpha_update`) does not invalidate the
+    capture."""
+
+    MAX_RECAPTURE = 4
+
+    def __init__(self, client):
+        args = client.args
+        self.client = client
+        self.args = args
+        self.ok = bool(
+            getattr(args, 'hip_graph', False) and torch.cuda.is_available()
+            and ops.hip_available() and not ops.FORCE_EAGER
+            and args.arch != 'rnn' and 'robust' not in args.arch)
+        self.graphs = None
+        self.sx = self.sy = None
+        self.alpha_t = None
+        self.lr0 = None
+        self.macc = None
+        self.pending = 0
+        self.recaptures = 0
+        self.fail = False
+        self.topk = None
+
+    def maybe_step(self, _input, _target, lr, alpha):
+        if not self.ok or self.fail or not _input.is_cuda:
+            return False
+        if self.graphs is not None:
+            if _input.shape != self.sx.shape or \
+                    _input.dtype != self.sx.dtype:
+                return self._eager_stolen(_input, _target, alpha)
+            if lr != self.lr0:
+                self.graphs = None
+                return self._eager_stolen(_input, _target, alpha)
+        if self.graphs is None:
+            if self.recaptures >= self.MAX_RECAPTURE:
+                self.fail = True
+                self._restore_eager()
+                return False
+            try:
+                self._capture(_input, _target, lr, alpha)
+            except Exception as e:  # noqa: BLE001
+                from fedtorch_amd.logs.logging import log
+                log('[graphstep-apfl] capture failed (%r)' % (e,),
+                    self.args.debug)
+                self.fail = True
+                self._restore_eager()
+                return False
+        self.sx.copy_(_input)
+        self.sy.copy_(_target)
+        self.alpha_t.fill_(float(alpha))
+        for g in self.graphs:
+            g.replay()
+        self.pending += 1
+        return True
+
+    def flush(self, tracker):
+        if self.pending == 0 or self.macc is None:
+            return
+        v = self.macc.cpu().tolist()
+        n = max(v[-1], 1.0)
+        tracker['losses'].update(v[0] / n, int(n))
+        names = ['top1', 'top5']
+        for i, k in enumerate(self.topk):
+            tracker[names[i]].update(100.0 * v[1 + i] / n, int(n))
+        self.macc.zero_()
+        self.pending = 0
+
+    # ------------------------------------------------------------------
+    def eager_step(self, x, y, alpha):
+        """Both model steps on the stolen-gather flow (twin-safe):
+        used for the adaptive-alpha step (which must read fresh grads
+        host-side right after) and odd batches."""
+        return self._eager_stolen(x, y, alpha)
+
+    def _eager_stolen(self, x, y, alpha):
+        c = self.client
+        args = self.args
+        c.arena.detach_grads()
+        c.arena_personal.detach_grads()
+        with amp(args):
+            loss = c.criterion(c.model(x), y)
+        loss.backward()
+        c.arena.gather_grads()
+        c.optimizer.step(apply_lr=True,
+                         apply_in_momentum=args.in_momentum,
+                         apply_out_momentum=False)
+        c.arena.detach_grads()
+        c.arena_personal.detach_grads()
+        with amp(args):
+            op = c.model_personal(x)
+            og = c.model(x)
+            out = (float(alpha) * op) + ((1.0 - float(alpha)) * og)
+            loss_p = c.criterion(out, y)
+        loss_p.backward()
+        c.arena_personal.gather_grads()
+        c.arena.gather_grads()
+        c.optimizer_personal.step(apply_lr=True,
+                                  apply_in_momentum=args.in_momentum,
+                                  apply_out_momentum=False)
+        if self.macc is not None and self.topk is not None:
+            with torch.no_grad():
+                od = out.detach().float()
+                _, pred = od.topk(max(self.topk), 1, True, True)
+                eq = pred.t().eq(y.view(1, -1))
+                self.macc[0] += loss_p.detach().float() * float(x.size(0))
+                for i, k in enumerate(self.topk):
+                    self.macc[1 + i] += eq[:k].sum().float()
+                self.macc[-1] += float(x.size(0))
+                self.pending += 1
+        return True
+
+    def _restore_eager(self):
+        c = self.client
+        for arena in (c.arena, c.arena_personal):
+            if arena.half_flat is not None:
+                arena.disable_bf16_compute()
+            if arena.grad is not None:
+                arena.attach_grads()
+
+    def _snap_arena(self, arena, opt):
+        s = {'flat': arena.flat.clone()}
+        if arena.buf_flat is not None:
+            s['buf'] = arena.buf_flat.clone()
+        if getattr(opt, '_in_buf', None) is not None:
+            s['mom'] = opt._in_buf.clone()
+            s['mom_init'] = opt._in_init
+        return s
+
+    def _restore_arena(self, arena, opt, s):
+        arena.flat.copy_(s['flat'])
+        if 'buf' in s:
+            arena.buf_flat.copy_(s['buf'])
+        if 'mom' in s:
+            opt._in_buf.copy_(s['mom'])
+            opt._in_init = s['mom_init']
+        elif getattr(opt, '_in_buf', None) is not None:
+            opt._in_buf.zero_()
+        if arena.half_flat is not None:
+            arena.sync_half()
+
+    def _capture(self, _input, _target, lr, alpha):
+        c = self.client
+        args = self.args
+        self.topk = tuple(c.metrics)
+        self.sx = _input.clone()
+        self.sy = _target.clone()
+        self.alpha_t = torch.full((), float(alpha), device=_input.device)
+        self.macc = torch.zeros(2 + len(self.topk), device=_input.device)
+        self.lr0 = lr
+        maxk = max(self.topk)
+        bs = float(_input.size(0))
+
+        if args.bf16:
+            if c.arena.half_flat is None:
+                c.arena.enable_bf16_compute()
+            if c.arena_personal.half_flat is None:
+                c.arena_personal.enable_bf16_compute()
+
+        snap_g = self._snap_arena(c.arena, c.optimizer)
+        snap_p = self._snap_arena(c.arena_personal, c.optimizer_personal)
+
+        def blend_fwd():
+            with amp(args):
+                op = c.model_personal(self.sx)
+                og = c.model(self.sx)
+                # match the eager float-scalar blend's rounding exactly:
+                # a 0-dim fp32 tensor would promote the bf16 logits to
+                # fp32 (graph==eager drifted 7e-2 over 8 steps)
+                out = (self.alpha_t * op).to(op.dtype) + \
+                    ((1.0 - self.alpha_t) * og).to(og.dtype)
+                loss = c.criterion(out, self.sy)
+            return out, loss
+
+        def inner():
+            c.arena.detach_grads()
+            c.arena_personal.detach_grads()
+            with amp(args):
+                loss = c.criterion(c.model(self.sx), self.sy)
+            loss.backward()
+            c.arena.gather_grads()
+            c.optimizer.step(apply_lr=True,
+                             apply_in_momentum=args.in_momentum,
+                             apply_out_momentum=False)
+            c.arena.detach_grads()
+            c.arena_personal.detach_grads()
+            _, loss_p = blend_fwd()
+            loss_p.backward()
+            c.arena_personal.gather_grads()
+            c.arena.gather_grads()  # blend also grads the global model
+            c.optimizer_personal.step(apply_lr=True,
+                                      apply_in_momentum=args.in_momentum,
+                                      apply_out_momentum=False)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                inner()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+
+        gs = []
+        c.arena.detach_grads()
+        c.arena_personal.detach_grads()
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            with amp(args):
+                loss = c.criterion(c.model(self.sx), self.sy)
+            loss.backward()
+        gs.append(g1)
+        c.arena.gather_grads()
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2):
+            c.arena.gather_grads()
+            c.optimizer.step(apply_lr=True,
+                             apply_in_momentum=args.in_momentum,
+                             apply_out_momentum=False)
+        gs.append(g2)
+        keep = [c.arena._gather_state]
+        c.arena.detach_grads()
+        c.arena_personal.detach_grads()
+        g3 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g3):
+            out, loss_p = blend_fwd()
+            loss_p.backward()
+            with torch.no_grad():
+                od = out.detach().float()
+                _, pred = od.topk(maxk, 1, True, True)
+                eq = pred.t().eq(self.sy.view(1, -1))
+                self.macc[0] += loss_p.detach().float() * bs
+                for i, k in enumerate(self.topk):
+                    self.macc[1 + i] += eq[:k].sum().float()
+                self.macc[-1] += bs
+        gs.append(g3)
+        c.arena_personal.gather_grads()
+        c.arena.gather_grads()
+        g4 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g4):
+            c.arena_personal.gather_grads()
+            c.arena.gather_grads()
+            c.optimizer_personal.step(apply_lr=True,
+                                      apply_in_momentum=args.in_momentum,
+                                      apply_out_momentum=False)
+        gs.append(g4)
+        keep.append(c.arena._gather_state)
+        keep.append(c.arena_personal._gather_state)
+        self._keep = keep
+        torch.cuda.synchronize()
+        self._restore_arena(c.arena, c.optimizer, snap_g)
+        self._restore_arena(c.arena_personal, c.optimizer_personal, snap_p)
+        self.macc.zero_()
+        self.graphs = gs
         self.recaptures += 1

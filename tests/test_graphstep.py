@@ -48,3 +48,46 @@ def test_hip_graph_matches_eager_on_gpu():
     diff = (a - b).abs().max().item()
     assert torch.allclose(a, b, atol=2e-3, rtol=1e-3), \
         'graph vs eager max diff %.2e' % diff
+
+
+def _run_apfl(hip_graph, on_cuda=False, seed=19):
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes import Client
+    from fedtorch_amd.trainings.apfl import train_and_validate_federated_apfl
+    argv = ['-d', 'mnist', '-a', 'cnn', '-f', 'true',
+            '--federated_type', 'apfl', '--fed_personal', 'true',
+            '--fed_personal_alpha', '0.4', '--num_comms', '2',
+            '--online_client_rate', '1.0',
+            '--federated_sync_type', 'local_step', '--local_step', '4',
+            '-b', '20', '--lr', '0.1', '--in_momentum', 'true',
+            '--on_cuda', 'true' if on_cuda else 'false',
+            '--bf16', 'true' if on_cuda else 'false',
+            '--hip_graph', 'true' if hip_graph else 'false',
+            '--debug', 'false', '-j', '0', '--manual_seed', str(seed),
+            '--checkpoint', '/tmp/ft_gsa_%d' % int(hip_graph)]
+    args = get_args(argv)
+    client = Client(args, 0)
+    client.initialize()
+    client.initialize_dataset()
+    client.load_local_dataset()
+    client.gen_aux_models()
+    train_and_validate_federated_apfl(client)
+    return (client.arena.clone_flat().float().cpu(),
+            client.arena_personal.clone_flat().float().cpu())
+
+
+def test_apfl_hip_graph_flag_inert_on_cpu():
+    a = _run_apfl(True)
+    b = _run_apfl(False)
+    assert torch.equal(a[0], b[0]) and torch.equal(a[1], b[1])
+
+
+@pytest.mark.gpu
+def test_apfl_hip_graph_matches_eager_on_gpu():
+    a = _run_apfl(True, on_cuda=True)
+    b = _run_apfl(False, on_cuda=True)
+    for i in range(2):
+        d = (a[i] - b[i]).abs().max().item()
+        assert torch.allclose(a[i], b[i], atol=3e-3, rtol=2e-3), \
+            'model %d graph vs eager max diff %.2e' % (i, d)
