@@ -69,6 +69,10 @@ def train_and_validate_federated_drfa(client):
         k = int(k[0])
 
         set_round_correction(client)
+        gs = getattr(client, 'graph_stepper', None)
+        if gs is None and getattr(args, 'hip_graph', False):
+            from fedtorch_amd.trainings.graphstep import GraphStepper
+            gs = client.graph_stepper = GraphStepper(client)
         local_steps = 0
         lr = args.old_learning_rate
         if online:
@@ -90,6 +94,15 @@ def train_and_validate_federated_drfa(client):
                     if _input.size(0) == 1:
                         is_sync = is_sync_fed(args)
                         break
+                    if gs is not None and gs.maybe_step(_input, _target,
+                                                        lr):
+                        # the kth snapshot happens BETWEEN replays (above)
+                        # so the graph path preserves DRFA semantics
+                        tracker['start_load_time'] = time.time()
+                        is_sync = is_sync_fed(args)
+                        if is_sync:
+                            break
+                        continue
                     client.optimizer.zero_grad()
                     with amp(args):
                         loss, _ = inference(client.model, client.criterion,
@@ -105,6 +118,8 @@ def train_and_validate_federated_drfa(client):
         else:
             log('Offline in this round. Waiting on others to finish!',
                 args.debug)
+        if gs is not None:
+            gs.flush(tracker)
 
         do_validate(args, client.model, client.optimizer, client.criterion,
                     client.metrics, client.train_loader, None,

@@ -57,6 +57,11 @@ class GraphStepper(object):
         self.recaptures = 0
         self.fail = False
         self.topk = None
+        self.loss_ref = None
+
+    def last_loss(self):
+        """Last replayed step's loss (device scalar; host read syncs)."""
+        return self.loss_ref
 
     # ------------------------------------------------------------------
     def maybe_step(self, _input, _target, lr):
@@ -229,6 +234,10 @@ class GraphStepper(object):
                 for i, k in enumerate(self.topk):
                     self.macc[1 + i] += eq[:k].sum().float()
                 self.macc[-1] += bs
+        # in-pool loss tensor: stable address across replays, so the
+        # AFL-style per-round "last local loss" read is one float() after
+        # the round (a collective sync point follows anyway)
+        self.loss_ref = loss.detach()
         arena.gather_grads()  # build the chunk table outside capture
         g2 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g2):
