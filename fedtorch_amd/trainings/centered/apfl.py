@@ -18,7 +18,7 @@ from fedtorch_amd.logs.logging import (
 from fedtorch_amd.logs.meter import define_local_training_tracker
 
 
-def train_and_validate_apfl_centered(Clients, Server):
+def train_and_validate_apfl_centered(Clients, Server, validate=True):
     log('start training and validation of APFL in a centered way.')
     tracker = define_local_training_tracker()
     start_global_time = time.time()
@@ -43,16 +43,18 @@ def train_and_validate_apfl_centered(Clients, Server):
             C.args.rounds_comm = args.rounds_comm
             local_steps = 0
             is_sync = False
-            do_validate_centered(C.args, Server.model, Server.criterion,
-                                 Server.metrics, Server.optimizer,
-                                 C.train_loader, Server.global_val_tracker,
-                                 val=False)
-            if args.fed_personal:
+            if validate:
                 do_validate_centered(C.args, Server.model, Server.criterion,
                                      Server.metrics, Server.optimizer,
-                                     C.val_loader,
-                                     Server.global_personal_val_tracker,
-                                     val=True)
+                                     C.train_loader, Server.global_val_tracker,
+                                     val=False)
+            if args.fed_personal:
+                if validate:
+                    do_validate_centered(C.args, Server.model, Server.criterion,
+                                         Server.metrics, Server.optimizer,
+                                         C.val_loader,
+                                         Server.global_personal_val_tracker,
+                                         val=True)
             while not is_sync:
                 for _input, _target in C.train_loader:
                     local_steps += 1
@@ -95,19 +97,21 @@ def train_and_validate_apfl_centered(Clients, Server):
                     is_sync = is_sync_fed(C.args)
                     if is_sync:
                         break
-            do_validate_centered(C.args, C.model, C.criterion, C.metrics,
-                                 C.optimizer, C.train_loader,
-                                 Server.local_val_tracker, val=False,
-                                 personal=True,
-                                 model_personal=C.model_personal,
-                                 alpha=C.args.fed_personal_alpha)
-            if args.fed_personal:
+            if validate:
                 do_validate_centered(C.args, C.model, C.criterion, C.metrics,
-                                     C.optimizer, C.val_loader,
-                                     Server.local_personal_val_tracker,
-                                     val=True, personal=True,
+                                     C.optimizer, C.train_loader,
+                                     Server.local_val_tracker, val=False,
+                                     personal=True,
                                      model_personal=C.model_personal,
                                      alpha=C.args.fed_personal_alpha)
+            if args.fed_personal:
+                if validate:
+                    do_validate_centered(C.args, C.model, C.criterion, C.metrics,
+                                         C.optimizer, C.val_loader,
+                                         Server.local_personal_val_tracker,
+                                         val=True, personal=True,
+                                         model_personal=C.model_personal,
+                                         alpha=C.args.fed_personal_alpha)
             tracker['start_sync_time'] = time.time()
             args.global_index += 1
             logging_sync_time(tracker)
