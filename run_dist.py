@@ -85,6 +85,8 @@ def main(args):
         '--lr_warmup_epochs': 3,
         '--lr_decay': 1.01,
         '--bf16': args.bf16,
+        '--hip_graph': args.hip_graph,
+        '--channels_last': args.bf16,  # NHWC pairs with the bf16 path
     }
     cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
            '--nproc-per-node', str(args.num_clients),
@@ -142,6 +144,9 @@ if __name__ == '__main__':
     # reference `run_mpi.py:140`: TMPDIR override for dataset staging
     parser.add_argument('-td', '--tmp_dir', default='/tmp', type=str)
     parser.add_argument('--bf16', action='store_true')
+    parser.add_argument('--hip_graph', action='store_true',
+                        help='hipGraph-capture the local steps '
+                             '(trainings/graphstep.py)')
     parser.add_argument('--master_port', default=29500, type=int)
     parser.add_argument('--dry_run', action='store_true',
                         help='print the torchrun command and exit')
