@@ -85,7 +85,7 @@ def main(args):
         '--lr_warmup_epochs': 3,
         '--lr_decay': 1.01,
         '--bf16': args.bf16,
-        '--hip_graph': args.hip_graph,
+        '--hip_graph': getattr(args, 'hip_graph', False),
         '--channels_last': args.bf16,  # NHWC pairs with the bf16 path
     }
     cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
