@@ -103,8 +103,8 @@ def train_and_validate_federated_afl(client):
                     args.local_index += 1
                     args.local_data_seen += len(_target)
                     get_current_epoch(args)
-                    adjust_learning_rate(args, client.optimizer,
-                                         client.scheduler)
+                    lr = adjust_learning_rate(args, client.optimizer,
+                                              client.scheduler)
                     _input, _target = load_data_batch(args, _input, _target,
                                                       tracker)
                     if _input.size(0) == 1:

@@ -91,3 +91,57 @@ def test_apfl_hip_graph_matches_eager_on_gpu():
         d = (a[i] - b[i]).abs().max().item()
         assert torch.allclose(a[i], b[i], atol=3e-3, rtol=2e-3), \
             'model %d graph vs eager max diff %.2e' % (i, d)
+
+
+def _run_loop(fed_type, hip_graph, on_cuda, drfa=False, seed=23):
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes import Client
+    argv = ['-d', 'mnist', '-a', 'cnn', '-f', 'true',
+            '--federated_type', fed_type, '--num_comms', '2',
+            '--online_client_rate', '1.0',
+            '--federated_sync_type', 'local_step', '--local_step', '4',
+            '-b', '20', '--lr', '0.1', '--in_momentum', 'true',
+            '--on_cuda', 'true' if on_cuda else 'false',
+            '--bf16', 'true' if on_cuda else 'false',
+            '--hip_graph', 'true' if hip_graph else 'false',
+            '--debug', 'false', '-j', '0', '--manual_seed', str(seed),
+            '--checkpoint', '/tmp/ft_gl_%s_%d' % (fed_type, int(hip_graph))]
+    if drfa:
+        argv += ['--federated_drfa', 'true']
+    args = get_args(argv)
+    client = Client(args, 0)
+    client.initialize()
+    client.initialize_dataset()
+    client.load_local_dataset()
+    client.gen_aux_models()
+    if drfa:
+        from fedtorch_amd.trainings.drfa import (
+            train_and_validate_federated_drfa)
+        # single-rank DRFA: k broadcast + lambda machinery degenerate
+        import fedtorch_amd.trainings.drfa as dm
+        torch.manual_seed(seed)  # fix the k draw across arms
+        train_and_validate_federated_drfa(client)
+    else:
+        from fedtorch_amd.trainings.afl import (
+            train_and_validate_federated_afl)
+        train_and_validate_federated_afl(client)
+    return client.arena.clone_flat().float().cpu()
+
+
+@pytest.mark.gpu
+def test_drfa_hip_graph_matches_eager_on_gpu():
+    a = _run_loop('fedavg', True, True, drfa=True)
+    b = _run_loop('fedavg', False, True, drfa=True)
+    d = (a - b).abs().max().item()
+    assert torch.allclose(a, b, atol=3e-3, rtol=2e-3), \
+        'drfa graph vs eager diff %.2e' % d
+
+
+@pytest.mark.gpu
+def test_afl_hip_graph_matches_eager_on_gpu():
+    a = _run_loop('afl', True, True)
+    b = _run_loop('afl', False, True)
+    d = (a - b).abs().max().item()
+    assert torch.allclose(a, b, atol=3e-3, rtol=2e-3), \
+        'afl graph vs eager diff %.2e' % d
