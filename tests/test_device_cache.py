@@ -70,3 +70,16 @@ def test_not_cacheable_fallback():
         raise AssertionError('should have raised')
     except NotCacheable:
         pass
+
+
+def test_subset_of_partition_unwrap():
+    """fed_personal splits produce Subset(Partition(ArrayDataset)) —
+    the cache must compose the index chains."""
+    ds, x, y = _mk()
+    part = Partition(ds, list(range(10, 60)))          # base idx 10..59
+    sub = torch.utils.data.Subset(part, list(range(5, 15)))  # -> 15..24
+    ld = DeviceCachedLoader(sub, 4, seed=2, shuffle=False, device='cpu')
+    b0x, b0y = next(iter(ld))
+    assert torch.equal(b0x, x[15:19])
+    assert torch.equal(b0y, y[15:19])
+    assert sum(b.shape[0] for b, _ in ld) == 10
