@@ -101,8 +101,13 @@ class _Conv3x3BNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
-        dy = dy.contiguous(memory_format=_CL)
+        # pop the deferred-BN tag by the INCOMING tensor's ptr before any
+        # contiguous() copy could change it (a missed tag would silently
+        # skip the BN backward transform)
         tag = _BNBWD_TAGS.pop(dy.data_ptr(), None) if _BNBWD_TAGS else None
+        dy = dy.contiguous(memory_format=_CL)
+        if tag is None and _BNBWD_TAGS:
+            tag = _BNBWD_TAGS.pop(dy.data_ptr(), None)
         if tag is not None and _DGRAD_ENABLED:
             xbn, z, coefs, relu, dres = tag
             e = _empty(dy.device)
@@ -169,8 +174,10 @@ class _Conv3x3S2BNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy, _dpart):
         x, weight = ctx.saved_tensors
-        dy = dy.contiguous(memory_format=_CL)
         tag = _BNBWD_TAGS.pop(dy.data_ptr(), None) if _BNBWD_TAGS else None
+        dy = dy.contiguous(memory_format=_CL)
+        if tag is None and _BNBWD_TAGS:
+            tag = _BNBWD_TAGS.pop(dy.data_ptr(), None)
         if tag is not None:
             # a deferred BN backward landed on the stride-2 conv (no TRF
             # kernel here): materialize dy_conv = A*mask(dz) + B + D*x_bn
