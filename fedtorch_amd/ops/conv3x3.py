@@ -108,7 +108,9 @@ class _Conv3x3BNFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=_CL)
         if tag is None and _BNBWD_TAGS:
             tag = _BNBWD_TAGS.pop(dy.data_ptr(), None)
-        if tag is not None and _DGRAD_ENABLED:
+        # a live tag means the BN already SKIPPED its dx pass: always
+        # honor it (ignoring it would treat dz as dy_conv silently)
+        if tag is not None:
             xbn, z, coefs, relu, dres = tag
             e = _empty(dy.device)
             dx, dyc = ops._C.conv3x3_dgrad_bn(
