@@ -21,8 +21,20 @@ def set_online_clients_centered(args):
     return sorted(online.tolist())
 
 
+def _n_online(online_clients):
+    """Weight denominator, DISTRIBUTED semantics (`fedavg.py:17-27`):
+    client 0 doubles as the server and counts toward the denominator
+    even when it is not sampled (it joins with weight 0).  The
+    reference's centered mode divides by len(online) instead, so its two
+    modes disagree whenever client 0 sits out a round — the oracle
+    (tests/test_equivalence.py, partial participation) pins the centered
+    simulator to the real distributed behavior."""
+    return len(online_clients) if 0 in online_clients \
+        else len(online_clients) + 1
+
+
 def _weights(args, online_clients, lambda_weight):
-    n_online = len(online_clients)
+    n_online = _n_online(online_clients)
     if lambda_weight is None:
         return {o: 1.0 / n_online for o in online_clients}
     return {o: float(lambda_weight[o]) * args.graph.n_nodes / n_online
@@ -60,7 +72,7 @@ def fedgate_aggregation_centered(Clients, Server, online_clients, local_steps,
                                  lr, lambda_weight=None):
     """reference centered `fedgate.py:7-58`."""
     args = Server.args
-    n_online = len(online_clients)
+    n_online = _n_online(online_clients)
     w = (1.0 / n_online) if lambda_weight is None else None
     agg = Server.grad
     agg.zero_()
@@ -99,7 +111,7 @@ def scaffold_aggregation_centered(Clients, Server, online_clients,
                                   local_steps, lr, lambda_weight=None):
     """reference centered `scaffold.py:3-49`."""
     args = Server.args
-    n_online = len(online_clients)
+    n_online = _n_online(online_clients)
     agg = Server.grad
     agg.zero_()
     diff = Server.work.setdefault('diff', Server.arena.new_buffer())
@@ -137,7 +149,7 @@ def qsparse_aggregation_centered(Clients, Server, online_clients, local_steps,
                                  lr, lambda_weight=None):
     """reference centered `qsparse.py:5-47`."""
     args = Server.args
-    n_online = len(online_clients)
+    n_online = _n_online(online_clients)
     agg = Server.grad
     agg.zero_()
     diff = Server.work.setdefault('diff', Server.arena.new_buffer())

@@ -220,3 +220,81 @@ def test_apfl_centered_equals_distributed():
     dp = float(np.abs(c_per - d_per).max())
     assert ds < 5e-5, 'server diff %.3e' % ds
     assert dp < 5e-5, 'personal diff %.3e' % dp
+
+
+def _centered_pp_worker(q):
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes.centered import ClientCentered, ServerCentered
+    from fedtorch_amd.trainings.centered.main import (
+        train_and_validate_federated_centered)
+    argv = [a if a != '1.0' else '0.5' for a in ARGV_COMMON] + [
+        '--federated_type', 'fedavg', '-j', '2',
+        '--checkpoint', '/tmp/ft_eq_c_pp']
+    args = get_args(argv)
+    import copy as _copy
+    import fedtorch_amd.components.dataset as _ds
+    _orig = _ds._make_loader
+
+    def _ml0(a, data, bs, sh, drop_last=False, tag=0):
+        a = _copy.copy(a)
+        a.num_workers = 0
+        a.pin_memory = False
+        return _orig(a, data, bs, sh, drop_last, tag)
+    _ds._make_loader = _ml0
+    Clients = {}
+    for i in range(2):
+        Clients[i] = ClientCentered(args, i) if i == 0 else \
+            ClientCentered(args, i, Partitioner=Clients[0].Partitioner)
+    Server = ServerCentered(Clients[0].args, Clients[0].model)
+    Server.enable_grad(Clients[0].train_loader)
+    train_and_validate_federated_centered(Clients, Server, validate=False)
+    q.put(Server.arena.clone_flat().numpy())
+
+
+def _dist_pp_worker(rank, q):
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = '29943'
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    dist.init_process_group('gloo', rank=rank, world_size=2)
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes import Client
+    from fedtorch_amd.trainings.federated import train_and_validate_federated
+    argv = [a if a != '1.0' else '0.5' for a in ARGV_COMMON] + [
+        '--federated_type', 'fedavg', '-j', '0',
+        '--checkpoint', '/tmp/ft_eq_d_pp']
+    args = get_args(argv)
+    client = Client(args, rank)
+    client.initialize()
+    client.initialize_dataset()
+    client.load_local_dataset()
+    client.gen_aux_models()
+    train_and_validate_federated(client, validate=False)
+    if rank == 0:
+        q.put(client.arena.clone_flat().numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_centered_equals_distributed_partial_participation():
+    """The oracle under online_client_rate < 1: the online-set sampling
+    (np.random on rank 0, broadcast) must line up with the centered
+    draw, and offline-round semantics must match across modes."""
+    ctx = mp.get_context('spawn')
+    qc = ctx.SimpleQueue()
+    pc = ctx.Process(target=_centered_pp_worker, args=(qc,))
+    pc.start()
+    qd = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_dist_pp_worker, args=(r, qd))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    c = torch.from_numpy(qc.get())
+    d = torch.from_numpy(qd.get())
+    pc.join(timeout=300)
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    diff = (c - d).abs().max().item()
+    assert torch.allclose(c, d, atol=5e-6, rtol=1e-5), \
+        'partial-participation max diff %.3e' % diff
