@@ -111,7 +111,11 @@ def train_and_validate_federated_packed(client, pack, validate=False):
             client.comm.broadcast(k_cut, src=0)
             k_cut = int(k_cut[0])
 
-        n_online = max(len(online), 1)
+        # weight denominator: DISTRIBUTED semantics — global client 0
+        # (the server) counts even when not sampled (`fedavg.py:17-27`,
+        # same rule as aggregation/federated.rank_weight)
+        n_online = len(online) if 0 in online else len(online) + 1
+        n_online = max(n_online, 1)
         weights = []
         for j in range(pack.C):
             gid = pack.global_id(j)
