@@ -83,13 +83,32 @@ class ClientPack(object):
         return self.rank * self.C + j
 
     def build_loaders(self):
-        """per-virtual-client data partitions over W*C clients."""
+        """per-virtual-client data partitions over W*C clients.
+
+        ONE partitioner is built (by the first local client) and shared
+        by the rank's other virtual clients — building one per client
+        gave every client a DIFFERENT index permutation (only the
+        global-client-0 call shuffles before the broadcast), so chunks
+        could OVERLAP across virtual clients.  Sharing mirrors the
+        centered mode (`main_centered.py:20-25`) and keeps the broadcast
+        count identical on every rank (one per rank, at j == 0)."""
         from fedtorch_amd.components.dataset import define_dataset
+        per_client_data = self.args.data in ('emnist', 'emnist_full',
+                                             'synthetic', 'shakespeare')
+        part = None
         for j in range(self.C):
             args_j = copy(self.args)
             args_j.graph = VirtualGraph(self.args.graph, self.global_id(j),
                                         self.total_clients)
-            loaders = define_dataset(args_j, shuffle=True, test=False)
+            if per_client_data:
+                loaders = define_dataset(args_j, shuffle=True, test=False)
+            elif part is None:
+                loaders, part = define_dataset(args_j, shuffle=True,
+                                               test=False,
+                                               return_partitioner=True)
+            else:
+                loaders = define_dataset(args_j, shuffle=True, test=False,
+                                         Partitioner=part)
             self.train_loaders[j] = loaders[0]
         # counters derived for the last one apply to all (equal splits)
         self.args.num_batches_train_per_device_per_epoch = \

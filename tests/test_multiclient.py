@@ -144,3 +144,46 @@ def test_partial_buffers_weighted_by_online_count():
     p1.partial_buffers([], total_online)
     assert torch.equal(p1.base.arena.buf_flat,
                        torch.zeros_like(p1.base.arena.buf_flat))
+
+
+def test_build_loaders_disjoint_partitions():
+    """Every virtual client must get a DISJOINT chunk of the dataset:
+    one shared partitioner per rank (building one per client gave each a
+    different permutation — chunks could overlap)."""
+    import os
+    import types
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes import Client
+    args = get_args([
+        '-d', 'mnist', '-a', 'logistic_regression', '-f', 'true',
+        '--federated_type', 'fedavg', '--num_comms', '1',
+        '--clients_per_rank', '4', '-b', '10', '--lr', '0.1',
+        '--on_cuda', 'false', '-j', '0', '--manual_seed', '7',
+        '--checkpoint', '/tmp/ft_dl', '--debug', 'false'])
+    client = Client(args, 0)
+    client.initialize()
+    client.initialize_dataset()
+    client.load_local_dataset()
+    client.gen_aux_models()
+    pack = ClientPack(client, 4)
+    pack.build_loaders()
+    types  # noqa: B018
+    idx_sets = []
+    for j in range(4):
+        ds = pack.train_loaders[j].dataset
+        base, idx = ds, None
+        # unwrap Partition layers to base indices
+        chain = []
+        while hasattr(base, 'indices') and hasattr(base, 'data'):
+            chain.append(list(base.indices))
+            base = base.data
+        # compose
+        ids = chain[-1]
+        for lvl in reversed(chain[:-1]):
+            ids = [ids[i] for i in lvl]
+        idx_sets.append(set(ids))
+    union = set()
+    for s_ in idx_sets:
+        assert not (union & s_), 'virtual-client chunks overlap!'
+        union |= s_
