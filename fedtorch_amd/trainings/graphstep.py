@@ -100,7 +100,7 @@ class GraphStepper(object):
         self.pending += 1
         return True
 
-    def flush(self, tracker):
+    def flush(self, tracker, lr=None):
         """Fold the in-graph metric accumulator into the tracker meters
         (called at the round's sync point, where a device sync happens
         anyway)."""
@@ -112,6 +112,8 @@ class GraphStepper(object):
         names = ['top1', 'top5']
         for i, k in enumerate(self.topk):
             tracker[names[i]].update(100.0 * v[1 + i] / n, int(n))
+        if lr is not None:
+            tracker['learning_rate'].update(lr, self.pending)
         self.macc.zero_()
         self.pending = 0
 

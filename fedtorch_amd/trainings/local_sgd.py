@@ -34,6 +34,10 @@ def train_and_validate(client):
     tracker['start_load_time'] = time.time()
     log('enter the training.', args.debug)
     args.comm_time.append(0.0)
+    gs = getattr(client, 'graph_stepper', None)
+    if gs is None and getattr(args, 'hip_graph', False):
+        from fedtorch_amd.trainings.graphstep import GraphStepper
+        gs = client.graph_stepper = GraphStepper(client)
 
     while True:
         for _input, _target in client.train_loader:
@@ -46,17 +50,21 @@ def train_and_validate(client):
             lr = adjust_learning_rate(args, client.optimizer,
                                       client.scheduler)
             _input, _target = load_data_batch(args, _input, _target, tracker)
-            client.optimizer.zero_grad()
-            with amp(args):
-                loss, performance = inference(
-                    client.model, client.criterion, client.metrics,
-                    _input, _target, rnn=args.arch == 'rnn')
-            loss.backward()
-            client.optimizer.step(
-                apply_lr=True,
-                apply_in_momentum=args.in_momentum,
-                apply_out_momentum=False)
-            logging_computing(tracker, loss, performance, _input, lr)
+            if gs is not None and _input.size(0) > 1 and \
+                    gs.maybe_step(_input, _target, lr):
+                pass  # hipGraph replay; metrics land at the sync flush
+            else:
+                client.optimizer.zero_grad()
+                with amp(args):
+                    loss, performance = inference(
+                        client.model, client.criterion, client.metrics,
+                        _input, _target, rnn=args.arch == 'rnn')
+                loss.backward()
+                client.optimizer.step(
+                    apply_lr=True,
+                    apply_in_momentum=args.in_momentum,
+                    apply_out_momentum=False)
+                logging_computing(tracker, loss, performance, _input, lr)
 
             is_sync = args.local_index % local_step == 0
             if args.epoch_ % 1 == 0:
@@ -64,6 +72,8 @@ def train_and_validate(client):
 
             if is_sync:
                 log('Enter synching', args.debug)
+                if gs is not None:
+                    gs.flush(tracker, lr=lr)
                 args.global_index += 1
                 aggregate_gradients(args, client.comm, client.arena,
                                     client.model_server, client.optimizer,
@@ -75,6 +85,8 @@ def train_and_validate(client):
                 start_global_time = time.time()
 
             if args.finish_one_epoch:
+                if gs is not None:
+                    gs.flush(tracker, lr=lr)
                 if args.epoch % args.eval_freq == 0 and \
                         args.graph.rank == 0:
                     do_validate(args, client.model, client.optimizer,

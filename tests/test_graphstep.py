@@ -145,3 +145,31 @@ def test_afl_hip_graph_matches_eager_on_gpu():
     d = (a - b).abs().max().item()
     assert torch.allclose(a, b, atol=3e-3, rtol=2e-3), \
         'afl graph vs eager diff %.2e' % d
+
+
+def test_local_sgd_hip_graph_flag_inert_on_cpu():
+    """plain distributed local SGD loop with --hip_graph on CPU: inert."""
+    os.environ['FEDTORCH_SYNTH_SIZE'] = '200'
+    from fedtorch_amd.parameters import get_args
+    from fedtorch_amd.nodes import Client
+    from fedtorch_amd.trainings.local_sgd import train_and_validate
+
+    def run(hg):
+        argv = ['-d', 'mnist', '-a', 'cnn', '-f', 'false',
+                '--stop_criteria', 'epoch', '--num_epochs', '1',
+                '--local_step', '4', '-b', '20', '--lr', '0.1',
+                '--in_momentum', 'true', '--on_cuda', 'false',
+                '--hip_graph', 'true' if hg else 'false',
+                '--debug', 'false', '-j', '0', '--manual_seed', '29',
+                '--checkpoint', '/tmp/ft_ls_%d' % int(hg)]
+        args = get_args(argv)
+        client = Client(args, 0)
+        client.initialize()
+        client.initialize_dataset()
+        client.load_local_dataset()
+        client.gen_aux_models()
+        train_and_validate(client)
+        return client.arena.clone_flat().float().cpu()
+    a = run(True)
+    b = run(False)
+    assert torch.equal(a, b)
