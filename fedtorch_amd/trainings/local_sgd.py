@@ -52,7 +52,10 @@ def train_and_validate(client):
             _input, _target = load_data_batch(args, _input, _target, tracker)
             if gs is not None and _input.size(0) > 1 and \
                     gs.maybe_step(_input, _target, lr):
-                pass  # hipGraph replay; metrics land at the sync flush
+                # hipGraph replay; metrics land at the sync flush — keep
+                # the timing bookkeeping logging_computing would have set
+                tracker['start_sync_time'] = time.time()
+                tracker['start_load_time'] = time.time()
             else:
                 client.optimizer.zero_grad()
                 with amp(args):
