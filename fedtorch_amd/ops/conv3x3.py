@@ -1,16 +1,23 @@
 # -*- coding: utf-8 -*-
-"""NHWC 3x3/s1/p1 conv with MFMA weight-gradient (`hip/convwrw.h`).
+"""NHWC 3x3 convolution modules on the MFMA kernel pack (round 2).
 
-Forward and the data gradient stay on MIOpen's native-NHWC igemm solvers
-(already fast); the WEIGHT gradient — whose MIOpen solvers carry
-SubTensorOp workspace-zero and fp32→bf16 cast wrapper kernels (~280 us of
-a 1.65 ms ResNet-20/b256 step, profiles/r01_bench_notes.md) — runs a
-hand-written MFMA 16x16x32 kernel computing all 9 taps as tile-GEMMs over
-the flattened position axis, no workspace, bf16 out.
+Default GPU training path for the CIFAR ResNet body convs (Co == Ci,
+(Co, W) in {(16,32), (32,16), (64,8)}) and the stride-2 transitions:
 
-Eligible shapes (compiled template instances): Co == Ci with
-(Co, W) in {(16,32), (32,16), (64,8)} — the CIFAR ResNet body convs.
-Anything else falls back to stock F.conv2d autograd.
+* forward: `hip/convfwd.h` conv3x3_bn_fwd_k — direct implicit GEMM with
+  a fused BN-stats epilogue (the following BN skips its stats pass via
+  the `_ft_bn_part` handoff) — 5.1-6.6 us/call;
+* backward-data: conv3x3_dgrad_k (the forward's mirror) — 5.0-6.7 us
+  vs MIOpen's 23-24;
+* weight gradient: MFMA wrw v2 (`hip/convwrw2.h`) for C16 where it
+  beats MIOpen-incl-wrappers (23.4 vs 33 us); MIOpen elsewhere;
+* deferred BN backward (opt-in): the BN's elementwise dx pass runs
+  inside the dgrad staging, handed off through a data_ptr-keyed side
+  table (python tensor attrs do not survive the autograd engine's
+  rewrapping).
+
+Anything else falls back to stock F.conv2d autograd.  Per-call evidence:
+profiles/r02_bench_notes.md; kernel details: hip/README.md.
 """
 import os
 

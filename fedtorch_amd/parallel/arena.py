@@ -236,7 +236,13 @@ class Arena(object):
 
     def gather_grads(self):
         """One-kernel copy of the stolen per-parameter grads into the grad
-        arena.  Builds the chunk table on first use (static shapes)."""
+        arena.  Builds the chunk table on first use (static shapes).
+
+        hipGraph captures BAKE the current table tensors' addresses into
+        the recorded kernel: the capturing code must keep a reference to
+        `_gather_state` for the graph's lifetime (see
+        trainings/graphstep.py `_keep`) — a later detach_grads() drops it
+        here and the memory can be reused."""
         if getattr(self, '_gather_state', None) is None:
             from fedtorch_amd import ops
             state = None
