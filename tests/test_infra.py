@@ -281,3 +281,26 @@ def test_reference_flag_surface_complete():
     ours = set(vars(args).keys())
     missing = sorted(set(REF_FLAGS) - ours)
     assert not missing, 'reference flags missing: %s' % missing
+
+
+def test_client_api_surface():
+    """The north star pins the reference's public Client API
+    (README.md:59-71: initialize / initialize_dataset /
+    load_local_dataset / gen_aux_models) and get_args."""
+    from fedtorch_amd.nodes import Client
+    from fedtorch_amd import parameters
+    for m in ('initialize', 'initialize_dataset', 'load_local_dataset',
+              'gen_aux_models'):
+        assert callable(getattr(Client, m, None)), m
+    assert callable(parameters.get_args)
+    from fedtorch_amd.trainings.federated import train_and_validate_federated
+    from fedtorch_amd.trainings.local_sgd import train_and_validate
+    from fedtorch_amd.trainings.drfa import (
+        train_and_validate_federated_drfa)
+    from fedtorch_amd.trainings.apfl import (
+        train_and_validate_federated_apfl)
+    from fedtorch_amd.trainings.afl import train_and_validate_federated_afl
+    assert all(callable(f) for f in (
+        train_and_validate_federated, train_and_validate,
+        train_and_validate_federated_drfa, train_and_validate_federated_apfl,
+        train_and_validate_federated_afl))
