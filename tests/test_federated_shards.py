@@ -116,3 +116,16 @@ def test_pipeline_shakespeare_real_shards(tmp_path):
         data_dir=str(tmp_path), rnn_seq_len=50)
     ds = get_dataset(args, 'shakespeare', str(tmp_path), split='train')
     assert isinstance(ds, fs.ShakespeareShards)
+
+
+def test_h5_adapter_raises_cleanly_without_h5py():
+    """no h5py in this image: the TFF adapter must fail with a clear
+    message, not an ImportError traceback mid-pipeline."""
+    import pytest as _pytest
+    try:
+        import h5py  # noqa: F401
+        _pytest.skip('h5py present')
+    except ImportError:
+        pass
+    with _pytest.raises(RuntimeError, match='h5py'):
+        fs.TFFClientH5('/tmp/nonexistent.h5')
